@@ -1,0 +1,292 @@
+"""FakeApiServer — in-memory kube-apiserver double.
+
+Faithful enough for the whole culler path: namespaced GETs for Pods and the
+five scalable kinds, RFC 7386 merge-PATCH on objects, the ``/scale``
+subresource, and Event POSTs. Objects are plain dicts; builder helpers mirror
+the reference's test fixture builders (reference lib.rs:590-652) plus pods
+with owner references for e2e-style flows.
+"""
+
+from __future__ import annotations
+
+import copy
+import json
+import re
+import threading
+import time
+import uuid
+from datetime import datetime, timedelta, timezone
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+# (api_prefix, plural) → kind
+_ROUTES = {
+    ("api/v1", "pods"): "Pod",
+    ("api/v1", "events"): "Event",
+    ("apis/apps/v1", "deployments"): "Deployment",
+    ("apis/apps/v1", "replicasets"): "ReplicaSet",
+    ("apis/apps/v1", "statefulsets"): "StatefulSet",
+    ("apis/kubeflow.org/v1", "notebooks"): "Notebook",
+    ("apis/serving.kserve.io/v1beta1", "inferenceservices"): "InferenceService",
+}
+
+_PATH_RE = re.compile(
+    r"^/(api/v1|apis/apps/v1|apis/kubeflow\.org/v1|apis/serving\.kserve\.io/v1beta1)"
+    r"/namespaces/([^/]+)/([^/]+)(?:/([^/]+))?(?:/(scale))?$"
+)
+
+
+def _now_rfc3339(offset_s: float = 0.0) -> str:
+    dt = datetime.now(timezone.utc) + timedelta(seconds=offset_s)
+    return dt.strftime("%Y-%m-%dT%H:%M:%SZ")
+
+
+def _merge_patch(target, patch):
+    if not isinstance(patch, dict) or not isinstance(target, dict):
+        return copy.deepcopy(patch)
+    for k, v in patch.items():
+        if v is None:
+            target.pop(k, None)
+        elif isinstance(v, dict) and isinstance(target.get(k), dict):
+            target[k] = _merge_patch(target[k], v)
+        else:
+            target[k] = copy.deepcopy(v)
+    return target
+
+
+class FakeApiServer:
+    def __init__(self, host: str = "127.0.0.1", port: int = 0, token: str | None = None,
+                 latency_s: float = 0.0):
+        self._lock = threading.Lock()
+        # objects[(kind, namespace, name)] = dict
+        self.objects: dict[tuple[str, str, str], dict] = {}
+        self.events: list[dict] = []
+        self.requests: list[tuple[str, str]] = []  # (method, path)
+        self.token = token
+        self.latency_s = latency_s  # simulated apiserver RTT for benchmarks
+
+        fixture = self
+
+        class Handler(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+            disable_nagle_algorithm = True
+
+            def log_message(self, *args):
+                pass
+
+            def _send(self, status: int, obj):
+                body = json.dumps(obj).encode()
+                self.send_response(status)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+            def _auth_ok(self) -> bool:
+                if fixture.token is None:
+                    return True
+                auth = self.headers.get("Authorization", "")
+                return auth == f"Bearer {fixture.token}"
+
+            def _route(self):
+                m = _PATH_RE.match(self.path.split("?")[0])
+                if not m:
+                    return None
+                prefix, ns, plural, name, sub = m.groups()
+                kind = _ROUTES.get((prefix, plural))
+                if kind is None:
+                    return None
+                return kind, ns, name, sub
+
+            def do_GET(self):
+                with fixture._lock:
+                    fixture.requests.append(("GET", self.path))
+                if fixture.latency_s:
+                    time.sleep(fixture.latency_s)
+                if not self._auth_ok():
+                    return self._send(401, {"kind": "Status", "code": 401})
+                r = self._route()
+                if r is None:
+                    return self._send(404, {"kind": "Status", "code": 404})
+                kind, ns, name, _sub = r
+                if name is None:
+                    with fixture._lock:
+                        items = [copy.deepcopy(o) for (k, n, _), o in fixture.objects.items()
+                                 if k == kind and n == ns]
+                    return self._send(200, {"kind": kind + "List", "items": items})
+                with fixture._lock:
+                    obj = fixture.objects.get((kind, ns, name))
+                    if obj is None:
+                        return self._send(404, {"kind": "Status", "code": 404,
+                                                "message": f"{kind} {ns}/{name} not found"})
+                    return self._send(200, copy.deepcopy(obj))
+
+            def do_PATCH(self):
+                with fixture._lock:
+                    fixture.requests.append(("PATCH", self.path))
+                if fixture.latency_s:
+                    time.sleep(fixture.latency_s)
+                if not self._auth_ok():
+                    return self._send(401, {"kind": "Status", "code": 401})
+                length = int(self.headers.get("Content-Length", "0"))
+                patch = json.loads(self.rfile.read(length) or b"{}")
+                r = self._route()
+                if r is None or r[2] is None:
+                    return self._send(404, {"kind": "Status", "code": 404})
+                kind, ns, name, sub = r
+                with fixture._lock:
+                    obj = fixture.objects.get((kind, ns, name))
+                    if obj is None:
+                        return self._send(404, {"kind": "Status", "code": 404})
+                    if sub == "scale":
+                        # /scale only understands spec.replicas
+                        replicas = (patch.get("spec") or {}).get("replicas")
+                        if replicas is None:
+                            return self._send(400, {"kind": "Status", "code": 400,
+                                                    "message": "scale patch needs spec.replicas"})
+                        obj.setdefault("spec", {})["replicas"] = replicas
+                        scale = {
+                            "kind": "Scale", "apiVersion": "autoscaling/v1",
+                            "metadata": {"name": name, "namespace": ns},
+                            "spec": {"replicas": replicas},
+                        }
+                        return self._send(200, scale)
+                    _merge_patch(obj, patch)
+                    return self._send(200, copy.deepcopy(obj))
+
+            def do_POST(self):
+                with fixture._lock:
+                    fixture.requests.append(("POST", self.path))
+                if fixture.latency_s:
+                    time.sleep(fixture.latency_s)
+                if not self._auth_ok():
+                    return self._send(401, {"kind": "Status", "code": 401})
+                length = int(self.headers.get("Content-Length", "0"))
+                obj = json.loads(self.rfile.read(length) or b"{}")
+                r = self._route()
+                if r is None:
+                    return self._send(404, {"kind": "Status", "code": 404})
+                kind, ns, _name, _sub = r
+                if kind == "Event":
+                    with fixture._lock:
+                        fixture.events.append(obj)
+                    return self._send(201, obj)
+                name = obj.get("metadata", {}).get("name", "")
+                with fixture._lock:
+                    fixture.objects[(kind, ns, name)] = obj
+                return self._send(201, obj)
+
+        self._server = ThreadingHTTPServer((host, port), Handler)
+        self._server.daemon_threads = True
+        self._thread = threading.Thread(
+            target=lambda: self._server.serve_forever(poll_interval=0.05), daemon=True)
+
+    # -- lifecycle -----------------------------------------------------------
+    def start(self) -> "FakeApiServer":
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._server.shutdown()
+        self._server.server_close()
+
+    @property
+    def url(self) -> str:
+        host, port = self._server.server_address[:2]
+        return f"http://{host}:{port}"
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()
+
+    # -- object builders -----------------------------------------------------
+    def put(self, kind: str, obj: dict):
+        meta = obj["metadata"]
+        with self._lock:
+            self.objects[(kind, meta["namespace"], meta["name"])] = obj
+        return obj
+
+    def get(self, kind: str, ns: str, name: str) -> dict | None:
+        with self._lock:
+            obj = self.objects.get((kind, ns, name))
+            return copy.deepcopy(obj) if obj is not None else None
+
+    @staticmethod
+    def _meta(name, ns, uid=None, owners=None, labels=None, age_s: float = 0.0):
+        meta = {
+            "name": name,
+            "namespace": ns,
+            "uid": uid or str(uuid.uuid4()),
+            "resourceVersion": "1",
+            "creationTimestamp": _now_rfc3339(-age_s),
+        }
+        if owners:
+            meta["ownerReferences"] = owners
+        if labels:
+            meta["labels"] = labels
+        return meta
+
+    def add_deployment(self, name, ns, uid=None, replicas=1):
+        return self.put("Deployment", {
+            "apiVersion": "apps/v1", "kind": "Deployment",
+            "metadata": self._meta(name, ns, uid),
+            "spec": {"replicas": replicas},
+        })
+
+    def add_replicaset(self, name, ns, uid=None, owner=None, replicas=1):
+        owners = None
+        if owner is not None:
+            owners = [{"apiVersion": "apps/v1", "kind": "Deployment",
+                       "name": owner["metadata"]["name"], "uid": owner["metadata"]["uid"]}]
+        return self.put("ReplicaSet", {
+            "apiVersion": "apps/v1", "kind": "ReplicaSet",
+            "metadata": self._meta(name, ns, uid, owners),
+            "spec": {"replicas": replicas},
+        })
+
+    def add_statefulset(self, name, ns, uid=None, notebook_owner=None, replicas=1):
+        owners = None
+        if notebook_owner is not None:
+            owners = [{"apiVersion": "kubeflow.org/v1", "kind": "Notebook",
+                       "name": notebook_owner["metadata"]["name"],
+                       "uid": notebook_owner["metadata"]["uid"]}]
+        return self.put("StatefulSet", {
+            "apiVersion": "apps/v1", "kind": "StatefulSet",
+            "metadata": self._meta(name, ns, uid, owners),
+            "spec": {"replicas": replicas},
+        })
+
+    def add_notebook(self, name, ns, uid=None):
+        return self.put("Notebook", {
+            "apiVersion": "kubeflow.org/v1", "kind": "Notebook",
+            "metadata": self._meta(name, ns, uid),
+            "spec": {"template": None},
+        })
+
+    def add_inferenceservice(self, name, ns, uid=None, min_replicas=1):
+        return self.put("InferenceService", {
+            "apiVersion": "serving.kserve.io/v1beta1", "kind": "InferenceService",
+            "metadata": self._meta(name, ns, uid),
+            "spec": {"predictor": {"minReplicas": min_replicas}},
+        })
+
+    def add_pod(self, name, ns, owner_kind=None, owner_name=None, owner_uid=None,
+                labels=None, phase="Running", age_s: float = 7200.0,
+                creation_timestamp: str | None = None):
+        owners = None
+        if owner_kind is not None:
+            owners = [{"apiVersion": "apps/v1", "kind": owner_kind,
+                       "name": owner_name, "uid": owner_uid or str(uuid.uuid4())}]
+        meta = self._meta(name, ns, None, owners, labels, age_s or 0.0)
+        if creation_timestamp is not None:
+            meta["creationTimestamp"] = creation_timestamp
+        elif age_s is None:  # explicitly no creation timestamp (skip-path test)
+            meta.pop("creationTimestamp", None)
+        pod = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": meta,
+            "spec": {},
+            "status": {"phase": phase},
+        }
+        return self.put("Pod", pod)

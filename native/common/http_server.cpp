@@ -1,0 +1,169 @@
+#include "http_server.hpp"
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <mutex>
+#include <stdexcept>
+
+#include "strutil.hpp"
+
+namespace http {
+
+Server::Server(const std::string& bind_addr, uint16_t port, Handler handler)
+    : bind_addr_(bind_addr), port_(port), handler_(std::move(handler)) {}
+
+Server::~Server() { stop(); }
+
+void Server::start() {
+  listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+  int one = 1;
+  ::setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof one);
+  struct sockaddr_in addr {};
+  addr.sin_family = AF_INET;
+  addr.sin_port = htons(port_);
+  if (::inet_pton(AF_INET, bind_addr_.c_str(), &addr.sin_addr) != 1)
+    throw std::runtime_error("invalid bind address: " + bind_addr_);
+  if (::bind(listen_fd_, reinterpret_cast<struct sockaddr*>(&addr), sizeof addr) < 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+    throw std::runtime_error("bind to " + bind_addr_ + ":" + std::to_string(port_) +
+                             " failed: " + std::strerror(errno));
+  }
+  if (port_ == 0) {
+    socklen_t len = sizeof addr;
+    ::getsockname(listen_fd_, reinterpret_cast<struct sockaddr*>(&addr), &len);
+    port_ = ntohs(addr.sin_port);
+  }
+  if (::listen(listen_fd_, 128) < 0) throw std::runtime_error("listen() failed");
+  running_.store(true);
+  accept_thread_ = std::thread([this] { accept_loop(); });
+}
+
+void Server::stop() {
+  if (!running_.exchange(false)) return;
+  ::shutdown(listen_fd_, SHUT_RDWR);
+  ::close(listen_fd_);
+  if (accept_thread_.joinable()) accept_thread_.join();
+  std::lock_guard<std::mutex> lock(workers_mu_);
+  for (auto& w : workers_)
+    if (w.joinable()) w.join();
+  workers_.clear();
+}
+
+void Server::accept_loop() {
+  while (running_.load()) {
+    int fd = ::accept(listen_fd_, nullptr, nullptr);
+    if (fd < 0) {
+      if (!running_.load()) break;
+      continue;
+    }
+    int one = 1;
+    ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+    std::lock_guard<std::mutex> lock(workers_mu_);
+    // reap finished threads opportunistically
+    if (workers_.size() > 64) {
+      for (auto& w : workers_)
+        if (w.joinable()) w.join();
+      workers_.clear();
+    }
+    workers_.emplace_back([this, fd] { handle_conn(fd); });
+  }
+}
+
+void Server::handle_conn(int fd) {
+  struct timeval tv {30, 0};
+  ::setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+  ::setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
+
+  std::string buf;
+  char tmp[16384];
+  while (running_.load()) {
+    // ---- read request head ----
+    size_t hdr_end;
+    while ((hdr_end = buf.find("\r\n\r\n")) == std::string::npos) {
+      ssize_t r = ::recv(fd, tmp, sizeof tmp, 0);
+      if (r <= 0) {
+        ::close(fd);
+        return;
+      }
+      buf.append(tmp, static_cast<size_t>(r));
+      if (buf.size() > (1u << 20)) {
+        ::close(fd);
+        return;
+      }
+    }
+    ServerRequest req;
+    {
+      std::string head = buf.substr(0, hdr_end);
+      auto lines = strutil::split(head, '\n');
+      auto parts = strutil::split(strutil::trim(lines[0]), ' ');
+      if (parts.size() < 2) {
+        ::close(fd);
+        return;
+      }
+      req.method = parts[0];
+      std::string target = parts[1];
+      size_t q = target.find('?');
+      req.path = q == std::string::npos ? target : target.substr(0, q);
+      req.query = q == std::string::npos ? "" : target.substr(q + 1);
+      for (size_t i = 1; i < lines.size(); i++) {
+        std::string line = strutil::trim(lines[i]);
+        size_t colon = line.find(':');
+        if (colon != std::string::npos)
+          req.headers[strutil::lower(line.substr(0, colon))] =
+              strutil::trim(line.substr(colon + 1));
+      }
+    }
+    size_t content_len = 0;
+    if (auto it = req.headers.find("content-length"); it != req.headers.end())
+      content_len = std::strtoul(it->second.c_str(), nullptr, 10);
+    size_t body_start = hdr_end + 4;
+    while (buf.size() < body_start + content_len) {
+      ssize_t r = ::recv(fd, tmp, sizeof tmp, 0);
+      if (r <= 0) {
+        ::close(fd);
+        return;
+      }
+      buf.append(tmp, static_cast<size_t>(r));
+    }
+    req.body = buf.substr(body_start, content_len);
+    buf.erase(0, body_start + content_len);
+
+    ServerResponse resp;
+    try {
+      resp = handler_(req);
+    } catch (const std::exception& e) {
+      resp.status = 500;
+      resp.body = std::string("internal error: ") + e.what();
+    }
+
+    const char* reason = resp.status == 200   ? "OK"
+                         : resp.status == 404 ? "Not Found"
+                         : resp.status == 400 ? "Bad Request"
+                                              : "Status";
+    std::string out = "HTTP/1.1 " + std::to_string(resp.status) + " " + reason + "\r\n";
+    out += "Content-Type: " + resp.content_type + "\r\n";
+    out += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
+    out += "Connection: keep-alive\r\n\r\n";
+    out += resp.body;
+    size_t off = 0;
+    while (off < out.size()) {
+      ssize_t w = ::send(fd, out.data() + off, out.size() - off, MSG_NOSIGNAL);
+      if (w <= 0) {
+        ::close(fd);
+        return;
+      }
+      off += static_cast<size_t>(w);
+    }
+  }
+  ::close(fd);
+}
+
+}  // namespace http

@@ -1,0 +1,37 @@
+// main.cpp — gpu-pruner binary entry point (MI355X-native culler daemon).
+//
+// Shape mirrors the reference's main(): parse CLI, init
+// logging + OTLP, run the producer/consumer daemon.
+#include <cstdio>
+
+#include "../common/log.hpp"
+#include "config.hpp"
+#include "daemon.hpp"
+#include "otlp.hpp"
+
+int main(int argc, char** argv) {
+  std::vector<std::string> args(argv + 1, argv + argc);
+  pruner::CliResult cli = pruner::parse_cli(args);
+  if (cli.show_help) {
+    std::fputs(pruner::cli_help().c_str(), stdout);
+    return 0;
+  }
+  if (cli.error) {
+    std::fprintf(stderr, "error: %s\n\n%s", cli.error->c_str(), pruner::cli_help().c_str());
+    return 2;
+  }
+
+  logx::Format fmt = logx::Format::Default;
+  switch (cli.config.log_format) {
+    case pruner::LogFormatOpt::Json: fmt = logx::Format::Json; break;
+    case pruner::LogFormatOpt::Pretty: fmt = logx::Format::Pretty; break;
+    default: break;
+  }
+  logx::init(fmt);
+  otlp::init("gpu-pruner");
+
+  int rc = pruner::run_daemon(cli.config);
+
+  otlp::shutdown();
+  return rc;
+}

@@ -1,0 +1,213 @@
+#include "otlp.hpp"
+
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "../common/http.hpp"
+#include "../common/json.hpp"
+#include "../common/log.hpp"
+#include "../common/strutil.hpp"
+
+namespace otlp {
+
+namespace {
+
+struct FinishedSpan {
+  std::string name;
+  uint64_t start_ns;
+  uint64_t end_ns;
+};
+
+struct State {
+  std::atomic<bool> enabled{false};
+  std::atomic<bool> running{false};
+  std::atomic<uint64_t> delivered{0};
+  std::string endpoint;  // base, no trailing slash
+  std::string service_name;
+  std::string trace_id;  // one id per process run; spans are flat (no parent tracking)
+  std::mutex mu;
+  std::vector<FinishedSpan> spans;
+  std::thread exporter;
+  std::condition_variable cv;
+  bool stop = false;
+  int interval_ms = 5000;
+};
+
+State& state() {
+  static State s;
+  return s;
+}
+
+uint64_t now_unix_ns() {
+  return static_cast<uint64_t>(std::chrono::duration_cast<std::chrono::nanoseconds>(
+                                   std::chrono::system_clock::now().time_since_epoch())
+                                   .count());
+}
+
+std::string rand_hex(size_t bytes) {
+  std::string id = strutil::uuid4_simple();  // 32 hex chars
+  return id.substr(0, bytes * 2);
+}
+
+jsn::Value resource_json(const std::string& service_name) {
+  jsn::Value attr = jsn::Value::object();
+  attr["key"] = "service.name";
+  attr["value"] = jsn::Value::object();
+  attr["value"]["stringValue"] = service_name;
+  jsn::Value res = jsn::Value::object();
+  res["attributes"] = jsn::Value(jsn::Array{attr});
+  return res;
+}
+
+void post_json(const std::string& url, const jsn::Value& body) {
+  auto parsed = http::Url::parse(url);
+  if (!parsed) return;
+  http::ClientOptions opts;
+  opts.connect_timeout_ms = 2000;
+  opts.io_timeout_ms = 5000;
+  http::Client client(*parsed, opts);
+  http::Request req;
+  req.method = "POST";
+  req.path = parsed->path;
+  req.body = body.dump();
+  req.headers.emplace_back("Content-Type", "application/json");
+  http::Response resp = client.request(req);
+  if (resp.status >= 200 && resp.status < 300)
+    state().delivered.fetch_add(1, std::memory_order_relaxed);
+}
+
+void export_once() {
+  State& s = state();
+  // ---- spans ----
+  std::vector<FinishedSpan> batch;
+  {
+    std::lock_guard<std::mutex> lock(s.mu);
+    batch.swap(s.spans);
+  }
+  if (!batch.empty()) {
+    jsn::Value spans = jsn::Value::array();
+    for (const auto& fs : batch) {
+      jsn::Value sp = jsn::Value::object();
+      sp["traceId"] = s.trace_id;
+      sp["spanId"] = rand_hex(8);
+      sp["name"] = fs.name;
+      sp["kind"] = 1;  // SPAN_KIND_INTERNAL
+      sp["startTimeUnixNano"] = std::to_string(fs.start_ns);
+      sp["endTimeUnixNano"] = std::to_string(fs.end_ns);
+      spans.push_back(sp);
+    }
+    jsn::Value scope_spans = jsn::Value::object();
+    scope_spans["scope"] = jsn::Value::object();
+    scope_spans["scope"]["name"] = "gpu_pruner::main";
+    scope_spans["spans"] = spans;
+    jsn::Value rs = jsn::Value::object();
+    rs["resource"] = resource_json(s.service_name);
+    rs["scopeSpans"] = jsn::Value(jsn::Array{scope_spans});
+    jsn::Value payload = jsn::Value::object();
+    payload["resourceSpans"] = jsn::Value(jsn::Array{rs});
+    try {
+      post_json(s.endpoint + "/v1/traces", payload);
+    } catch (const std::exception&) { /* collector away; drop batch */ }
+  }
+
+  // ---- metrics: the counter registry (monotonic counters + gauges) ----
+  auto counters = logx::counters_snapshot();
+  if (!counters.empty()) {
+    uint64_t t = now_unix_ns();
+    jsn::Value metrics = jsn::Value::array();
+    for (const auto& [name, value] : counters) {
+      jsn::Value dp = jsn::Value::object();
+      dp["asInt"] = std::to_string(value);
+      dp["timeUnixNano"] = std::to_string(t);
+      jsn::Value m = jsn::Value::object();
+      bool monotonic = strutil::starts_with(name, "monotonic_counter.");
+      std::string short_name = name.substr(name.find('.') + 1);
+      m["name"] = short_name;
+      if (monotonic) {
+        dp["startTimeUnixNano"] = std::to_string(t);
+        m["sum"] = jsn::Value::object();
+        m["sum"]["dataPoints"] = jsn::Value(jsn::Array{dp});
+        m["sum"]["aggregationTemporality"] = 2;  // CUMULATIVE
+        m["sum"]["isMonotonic"] = true;
+      } else {
+        m["gauge"] = jsn::Value::object();
+        m["gauge"]["dataPoints"] = jsn::Value(jsn::Array{dp});
+      }
+      metrics.push_back(m);
+    }
+    jsn::Value scope_metrics = jsn::Value::object();
+    scope_metrics["scope"] = jsn::Value::object();
+    scope_metrics["scope"]["name"] = "gpu_pruner::main";
+    scope_metrics["metrics"] = metrics;
+    jsn::Value rm = jsn::Value::object();
+    rm["resource"] = resource_json(s.service_name);
+    rm["scopeMetrics"] = jsn::Value(jsn::Array{scope_metrics});
+    jsn::Value payload = jsn::Value::object();
+    payload["resourceMetrics"] = jsn::Value(jsn::Array{rm});
+    try {
+      post_json(s.endpoint + "/v1/metrics", payload);
+    } catch (const std::exception&) { /* collector away */ }
+  }
+}
+
+}  // namespace
+
+void init(const std::string& service_name) {
+  State& s = state();
+  const char* disabled = std::getenv("OTEL_SDK_DISABLED");
+  if (disabled && strutil::lower(disabled) == "true") return;
+  const char* ep = std::getenv("OTEL_EXPORTER_OTLP_ENDPOINT");
+  if (!ep || !*ep) return;
+  s.endpoint = ep;
+  while (!s.endpoint.empty() && s.endpoint.back() == '/') s.endpoint.pop_back();
+  s.service_name = service_name;
+  if (const char* sn = std::getenv("OTEL_SERVICE_NAME"); sn && *sn) s.service_name = sn;
+  s.trace_id = rand_hex(16);
+  if (const char* iv = std::getenv("OTEL_METRIC_EXPORT_INTERVAL"); iv && *iv)
+    s.interval_ms = std::atoi(iv);
+  s.enabled.store(true);
+  s.running.store(true);
+  s.exporter = std::thread([&s] {
+    std::unique_lock<std::mutex> lock(s.mu);
+    while (!s.stop) {
+      s.cv.wait_for(lock, std::chrono::milliseconds(s.interval_ms));
+      if (s.stop) break;
+      lock.unlock();
+      export_once();
+      lock.lock();
+    }
+  });
+}
+
+void shutdown() {
+  State& s = state();
+  if (!s.running.load()) return;
+  {
+    std::lock_guard<std::mutex> lock(s.mu);
+    s.stop = true;
+  }
+  s.cv.notify_all();
+  s.exporter.join();
+  export_once();  // final flush
+  s.running.store(false);
+  s.enabled.store(false);
+}
+
+bool enabled() { return state().enabled.load(std::memory_order_relaxed); }
+
+SpanGuard::SpanGuard(const std::string& name) : name_(name), start_ns_(now_unix_ns()) {}
+
+SpanGuard::~SpanGuard() {
+  if (!enabled()) return;
+  State& s = state();
+  std::lock_guard<std::mutex> lock(s.mu);
+  s.spans.push_back({name_, start_ns_, now_unix_ns()});
+}
+
+uint64_t delivered_batches() { return state().delivered.load(std::memory_order_relaxed); }
+
+}  // namespace otlp

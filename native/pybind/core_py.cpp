@@ -1,0 +1,237 @@
+// core_py.cpp — _pruner_core: Python bindings over the native pruner core.
+//
+// Exposes the pure-logic layers (query builder, resource flags, scale-target
+// model, event generation, series parsing, CLI parsing) for the pytest suite
+// and bench harness. The daemon binaries do NOT go through Python — these
+// bindings exist so the unit-test surface of the reference
+// (gpu-pruner/src/lib.rs:578-998, main.rs:572-740) can be ported to pytest
+// against the very same native code the binaries link.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "../common/json.hpp"
+#include "../common/log.hpp"
+#include "../pruner/config.hpp"
+#include "../pruner/engine.hpp"
+#include "../pruner/k8s.hpp"
+#include "../pruner/prom.hpp"
+#include "../pruner/promql.hpp"
+#include "../pruner/resources.hpp"
+
+namespace py = pybind11;
+using namespace pruner;
+
+namespace {
+
+QueryArgs query_args_from_json(const std::string& json_args) {
+  jsn::Value v = jsn::parse(json_args);
+  QueryArgs qa;
+  if (v.contains("duration")) qa.duration_min = v.get("duration").as_int(30);
+  if (v.get("namespace").is_string()) qa.namespace_re = v.get("namespace").as_string();
+  if (v.get("model_name").is_string()) qa.model_name_re = v.get("model_name").as_string();
+  if (v.get("power_threshold").is_num())
+    qa.power_threshold_w = v.get("power_threshold").as_double();
+  qa.honor_labels = v.get("honor_labels").as_bool(false);
+  return qa;
+}
+
+Kind kind_from_str(const std::string& s) {
+  if (s == "Deployment") return Kind::Deployment;
+  if (s == "ReplicaSet") return Kind::ReplicaSet;
+  if (s == "StatefulSet") return Kind::StatefulSet;
+  if (s == "InferenceService") return Kind::InferenceService;
+  if (s == "Notebook") return Kind::Notebook;
+  throw py::value_error("unknown kind: " + s);
+}
+
+Config config_from_json(const std::string& json_cfg) {
+  jsn::Value v = jsn::parse(json_cfg);
+  Config c;
+  if (v.contains("duration")) c.duration_min = v.get("duration").as_int(30);
+  if (v.contains("grace_period")) c.grace_period_s = v.get("grace_period").as_int(300);
+  if (v.get("namespace").is_string()) c.namespace_ = v.get("namespace").as_string();
+  if (v.get("model_name").is_string()) c.model_name = v.get("model_name").as_string();
+  if (v.get("power_threshold").is_num())
+    c.power_threshold = v.get("power_threshold").as_double();
+  c.honor_labels = v.get("honor_labels").as_bool(false);
+  if (v.contains("max_concurrency"))
+    c.max_concurrency = static_cast<int>(v.get("max_concurrency").as_int(32));
+  if (v.get("run_mode").as_string_or("dry-run") == "scale-down")
+    c.run_mode = RunMode::ScaleDown;
+  c.enabled_resources = v.get("enabled_resources").as_string_or("drsin");
+  if (v.get("prometheus_url").is_string())
+    c.prometheus_url = v.get("prometheus_url").as_string();
+  return c;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_pruner_core, m) {
+  m.doc() = "native core of the MI355X gpu-pruner (query builder, scaling model, engine)";
+
+  logx::init(logx::Format::Default);
+
+  // ---- query builder ----
+  m.def("render_query", [](const std::string& json_args) {
+    return build_idle_query(query_args_from_json(json_args));
+  },
+      "Build the idle-GPU PromQL query from a JSON args object "
+      "(duration, namespace, model_name, power_threshold, honor_labels)");
+
+  // ---- resource flags ----
+  m.def("get_enabled_resources",
+        [](const std::string& s) { return static_cast<int>(get_enabled_resources(s)); });
+  m.attr("RK_DEPLOYMENT") = static_cast<int>(RK_DEPLOYMENT);
+  m.attr("RK_REPLICA_SET") = static_cast<int>(RK_REPLICA_SET);
+  m.attr("RK_STATEFUL_SET") = static_cast<int>(RK_STATEFUL_SET);
+  m.attr("RK_INFERENCE_SERVICE") = static_cast<int>(RK_INFERENCE_SERVICE);
+  m.attr("RK_NOTEBOOK") = static_cast<int>(RK_NOTEBOOK);
+
+  // ---- scale-target model ----
+  py::class_<ScaleKind>(m, "ScaleKind")
+      .def(py::init([](const std::string& kind, const std::string& obj_json) {
+             return ScaleKind{kind_from_str(kind), jsn::parse(obj_json)};
+           }),
+           py::arg("kind"), py::arg("object_json"))
+      .def_property_readonly("kind", [](const ScaleKind& sk) { return sk.kind_str(); })
+      .def_property_readonly("name", [](const ScaleKind& sk) { return sk.name(); })
+      .def_property_readonly("namespace",
+                             [](const ScaleKind& sk) -> py::object {
+                               auto ns = sk.ns();
+                               return ns ? py::cast(*ns) : py::none();
+                             })
+      .def_property_readonly("uid",
+                             [](const ScaleKind& sk) -> py::object {
+                               auto u = sk.uid();
+                               return u ? py::cast(*u) : py::none();
+                             })
+      .def_property_readonly("api_version",
+                             [](const ScaleKind& sk) { return sk.api_version(); })
+      .def_property_readonly("resource_version",
+                             [](const ScaleKind& sk) -> py::object {
+                               auto rv = sk.resource_version();
+                               return rv ? py::cast(*rv) : py::none();
+                             })
+      .def_property_readonly("resource_kind",
+                             [](const ScaleKind& sk) { return static_cast<int>(kind_flag(sk.kind)); })
+      .def("object_json", [](const ScaleKind& sk) { return sk.object.dump(); })
+      .def("__eq__", [](const ScaleKind& a, const ScaleKind& b) { return a == b; },
+           py::is_operator())
+      .def("__hash__", [](const ScaleKind& sk) { return sk.hash(); })
+      .def("__repr__", [](const ScaleKind& sk) {
+        return "<ScaleKind " + sk.kind_str() + " " + sk.ns().value_or("") + ":" + sk.name() + ">";
+      });
+
+  m.def("generate_scale_event",
+        [](const ScaleKind& sk) { return generate_scale_event(sk).dump(); });
+
+  // ---- series parsing ----
+  m.def("parse_pod_metric", [](const std::string& series_json) {
+    PodMetricData pmd = parse_pod_metric(jsn::parse(series_json));
+    py::dict d;
+    d["name"] = pmd.name;
+    d["namespace"] = pmd.ns;
+    d["container"] = pmd.container;
+    d["node_type"] = pmd.node_type;
+    d["gpu_model"] = pmd.gpu_model;
+    d["value"] = pmd.value;
+    return d;
+  });
+
+  py::register_exception<PodConvertError>(m, "PodConvertError");
+
+  // ---- CLI ----
+  m.def("parse_cli", [](const std::vector<std::string>& args) {
+    CliResult r = parse_cli(args);
+    py::dict d;
+    d["help"] = r.show_help;
+    d["error"] = r.error ? py::cast(*r.error) : py::object(py::none());
+    py::dict c;
+    c["duration"] = r.config.duration_min;
+    c["daemon_mode"] = r.config.daemon_mode;
+    c["enabled_resources"] = r.config.enabled_resources;
+    c["check_interval"] = r.config.check_interval_s;
+    c["namespace"] =
+        r.config.namespace_ ? py::cast(*r.config.namespace_) : py::object(py::none());
+    c["grace_period"] = r.config.grace_period_s;
+    c["model_name"] =
+        r.config.model_name ? py::cast(*r.config.model_name) : py::object(py::none());
+    c["power_threshold"] = r.config.power_threshold ? py::cast(*r.config.power_threshold)
+                                                    : py::object(py::none());
+    c["honor_labels"] = r.config.honor_labels;
+    c["run_mode"] = r.config.run_mode == RunMode::ScaleDown ? "scale-down" : "dry-run";
+    c["prometheus_url"] = r.config.prometheus_url;
+    c["prometheus_tls_mode"] =
+        r.config.prometheus_tls_mode == TlsModeOpt::Skip ? "skip" : "verify";
+    c["log_format"] = r.config.log_format == LogFormatOpt::Json      ? "json"
+                      : r.config.log_format == LogFormatOpt::Pretty ? "pretty"
+                                                                     : "default";
+    c["max_concurrency"] = r.config.max_concurrency;
+    c["queue_capacity"] = r.config.queue_capacity;
+    d["config"] = c;
+    return d;
+  });
+
+  // ---- engine (against a live endpoint: fake fixtures or a real cluster) ----
+  // Used by unit tests and bench.py; the daemon binary has its own main loop.
+  m.def("find_root_object",
+        [](const std::string& pod_json) -> py::object {
+          KubeClient kube(KubeConfig::resolve());
+          auto sk = find_root_object(kube, jsn::parse(pod_json));
+          if (!sk) return py::none();
+          return py::cast(*sk);
+        },
+        "Owner-reference walk using the env-configured apiserver");
+
+  m.def("scale", [](const ScaleKind& sk) {
+    KubeClient kube(KubeConfig::resolve());
+    scale(kube, sk);
+  });
+
+  m.def("evaluate_candidates",
+        [](const std::string& result_vector_json, const std::string& cfg_json) {
+          KubeClient kube(KubeConfig::resolve());
+          Config cfg = config_from_json(cfg_json);
+          QueryOutcome out;
+          std::vector<ScaleKind> roots =
+              evaluate_candidates(kube, jsn::parse(result_vector_json), cfg, &out);
+          py::gil_scoped_acquire gil;
+          py::dict d;
+          d["num_series"] = out.num_series;
+          d["num_unique_pods"] = out.num_unique_pods;
+          d["shutdown_events"] = out.shutdown_events;
+          py::list lst;
+          for (auto& sk : roots) lst.append(py::cast(sk));
+          d["roots"] = lst;
+          return d;
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "Full decision pass over a parsed Prometheus vector result");
+
+  m.def("run_tick",
+        [](const std::string& cfg_json) {
+          Config cfg = config_from_json(cfg_json);
+          auto prom = build_prom_client(cfg);
+          KubeClient kube(KubeConfig::resolve());
+          std::string query = build_idle_query(cfg.query_args());
+          QueryOutcome out = run_query_and_scale(*prom, kube, query, cfg, nullptr);
+          py::gil_scoped_acquire gil;
+          py::dict d;
+          d["num_series"] = out.num_series;
+          d["num_unique_pods"] = out.num_unique_pods;
+          d["shutdown_events"] = out.shutdown_events;
+          return d;
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "One full decision tick: Prometheus query + evaluation (dry-run unless "
+        "cfg.run_mode=scale-down, in which case scaling happens inline)");
+
+  m.def("get_prometheus_token", [] { return get_prometheus_token(); });
+
+  m.def("counters_snapshot", [] {
+    py::dict d;
+    for (const auto& [k, v] : logx::counters_snapshot()) d[py::str(k)] = v;
+    return d;
+  });
+  m.def("counters_reset", [] { logx::counters_reset_for_test(); });
+}
