@@ -1,0 +1,207 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark of the MI355X-native gpu-pruner.
+
+Measures the BASELINE.json north-star metric: **pods evaluated/sec** (and p50
+scale-decision latency) of the culler's decision engine on the synthetic
+1000-pod cluster, with the utilization signal read from the real MI355X GPU
+of each rank by the first-party ROCm sampler.
+
+One "step" = one full daemon tick in scale-down mode: Prometheus instant
+query → series parse → (pod,ns) dedup → concurrent pod GETs + eligibility
+filters → owner-reference walks → parent dedup → Event POST + scale PATCH for
+every selected root. Nothing is skipped inside the timed region; the fake
+Prometheus + apiserver are the native C++ synthetic backend (in-process HTTP
+over loopback), one instance per rank.
+
+Scaling is STRONG: 1000 pods total are split across the N ranks (BASELINE
+config 5: "1000 synthetic pods across 8 GPUs"); each rank monitors its own
+GPU via rocm_smi and feeds that activity value into its series.
+
+Usage:
+    python bench.py [--gpus N] [--steps K] [--warmup W] [--pods P]
+                    [--latency-us L] [--concurrency C]
+
+For N > 1 the driver launches this under torch.distributed.run with one rank
+per GPU; RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* are read from the env.
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO_ROOT)
+
+os.environ.setdefault("GPU_PRUNER_LOG", "error")
+
+TOTAL_PODS_DEFAULT = 1000
+
+
+def log(msg):
+    print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1,
+                    help="GPUs monitored (ranks when launched distributed)")
+    ap.add_argument("--steps", type=int, default=20, help="timed decision ticks")
+    ap.add_argument("--warmup", type=int, default=3, help="untimed warmup ticks")
+    ap.add_argument("--pods", type=int, default=TOTAL_PODS_DEFAULT,
+                    help="total synthetic pods across all ranks")
+    ap.add_argument("--latency-us", type=int, default=0,
+                    help="injected apiserver latency per request (RTT emulation)")
+    ap.add_argument("--concurrency", type=int,
+                    default=int(os.environ.get("BENCH_CONCURRENCY", "32")),
+                    help="engine --max-concurrency")
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world_size > 1
+
+    n_gpus = world_size if distributed else args.gpus
+    if distributed:
+        # Control-plane workload: no tensor collectives exist in this
+        # framework (SURVEY.md §5.8 — the reference has no data plane), so
+        # the barrier/reduce backend is gloo; CUDA work is still synchronized
+        # per-rank below.
+        dist.init_process_group(backend="gloo")
+
+    have_cuda = torch.cuda.is_available()
+    if have_cuda:
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+
+    from gpu_pruner_amd import _pruner_core as core
+
+    # ---- per-rank slice of the synthetic cluster (strong scaling) ----
+    pods_per_rank = max(1, args.pods // n_gpus)
+    backend = core.SyntheticBackend(n_pods=pods_per_rank, pods_per_parent=2,
+                                    gpus_per_pod=1, latency_us=args.latency_us,
+                                    model_name="AMD Instinct MI355X")
+    backend.start()
+    os.environ["GPU_PRUNER_K8S_URL"] = backend.k8s_url
+    os.environ["PROMETHEUS_TOKEN"] = "bench-token"
+    cfg = json.dumps({
+        "duration": 30, "grace_period": 300, "run_mode": "scale-down",
+        "prometheus_url": backend.prom_url, "max_concurrency": args.concurrency,
+        "model_name": "AMD Instinct MI355X",
+    })
+
+    # ---- real-GPU utilization feed (the rank's own device) ----
+    sampler = None
+    if have_cuda:
+        from gpu_pruner_amd import _gpumon
+        sampler = _gpumon.Sampler(poll_interval_ms=200)
+        sampler.init()  # raises loudly if rocm_smi/amdgpu is unavailable
+        sampler.start()
+        if rank == 0:
+            snap = sampler.snapshot()[0]
+            log(f"[bench] sampler: {len(sampler.snapshot())} GPU(s), "
+                f"model={snap['model_name']!r}")
+
+    my_device = local_rank
+    def feed_gpu_signal():
+        """Read the rank's GPU activity and feed it to the series source."""
+        if sampler is None:
+            backend.set_series_value(0.0)
+            return
+        snaps = sampler.snapshot()
+        d = snaps[my_device % len(snaps)]
+        backend.set_series_value(d["gr_engine_active"])
+
+    def one_step():
+        feed_gpu_signal()
+        out = core.run_tick(cfg)
+        return out
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        out = one_step()
+    expected = {"num_unique_pods": pods_per_rank}
+    if out["num_unique_pods"] != pods_per_rank:
+        log(f"[bench] WARNING rank {rank}: evaluated {out['num_unique_pods']} != "
+            f"{pods_per_rank} pods (GPU busy? series value nonzero)")
+
+    # ---- timed region ----
+    if distributed:
+        dist.barrier()
+    if have_cuda:
+        torch.cuda.synchronize()
+    step_times = []
+    t_start = time.perf_counter()
+    for _ in range(args.steps):
+        t0 = time.perf_counter()
+        out = one_step()
+        step_times.append(time.perf_counter() - t0)
+    if have_cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    t_end = time.perf_counter()
+
+    elapsed = t_end - t_start
+    # MAX elapsed over ranks (driver contract)
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        evaluated = torch.tensor([float(out["num_unique_pods"] * args.steps)],
+                                 dtype=torch.float64)
+        dist.all_reduce(evaluated, op=dist.ReduceOp.SUM)
+        total_evaluated = float(evaluated.item())
+    else:
+        total_evaluated = float(out["num_unique_pods"] * args.steps)
+
+    pods_per_sec = total_evaluated / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    p50 = statistics.median(step_times) * 1000.0
+    p95 = sorted(step_times)[max(0, int(len(step_times) * 0.95) - 1)] * 1000.0
+
+    backend.stop()
+    if sampler is not None:
+        sampler.stop()
+
+    if rank == 0:
+        result = {
+            "metric": "pods_evaluated_per_sec",
+            "value": round(pods_per_sec, 1),
+            "unit": "pods/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "n/a",
+            "data": "synthetic",
+            "config": {
+                "model": "gpu-pruner decision engine (idle-cull tick, scale-down mode)",
+                "n_pods": pods_per_rank * n_gpus,
+                "pods_per_rank": pods_per_rank,
+                "parallelism": f"rank-sharded x{n_gpus}",
+                "max_concurrency": args.concurrency,
+                "apiserver_latency_us": args.latency_us,
+                "p50_scale_decision_latency_ms": round(p50, 3),
+                "p95_scale_decision_latency_ms": round(p95, 3),
+                "events_posted": backend.events_posted,
+                "utilization_source": "rocm_smi sampler (real GPU)" if sampler else
+                                      "synthetic idle (no GPU)",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

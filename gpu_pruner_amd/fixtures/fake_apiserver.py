@@ -104,6 +104,20 @@ class FakeApiServer:
                     time.sleep(fixture.latency_s)
                 if not self._auth_ok():
                     return self._send(401, {"kind": "Status", "code": 401})
+                # cluster-scope pod list (used by the exporter's attribution
+                # cache), with optional spec.nodeName fieldSelector
+                path_only, _, query = self.path.partition("?")
+                if path_only == "/api/v1/pods":
+                    import urllib.parse as _up
+                    sel = _up.parse_qs(query).get("fieldSelector", [""])[0]
+                    node = None
+                    if sel.startswith("spec.nodeName="):
+                        node = sel.split("=", 1)[1]
+                    with fixture._lock:
+                        items = [copy.deepcopy(o) for (k, _, _), o in fixture.objects.items()
+                                 if k == "Pod" and (node is None or
+                                                   o.get("spec", {}).get("nodeName") == node)]
+                    return self._send(200, {"kind": "PodList", "items": items})
                 r = self._route()
                 if r is None:
                     return self._send(404, {"kind": "Status", "code": 404})

@@ -7,16 +7,51 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <chrono>
+#include <condition_variable>
 #include <cstring>
 #include <mutex>
+#include <set>
 #include <stdexcept>
 
 #include "strutil.hpp"
 
 namespace http {
 
+// Connection bookkeeping shared with detached handler threads: stop() must be
+// able to shut down every live keep-alive socket without joining threads that
+// are blocked in recv().
+struct ConnRegistry {
+  std::mutex mu;
+  std::condition_variable cv;
+  std::set<int> fds;
+  int active = 0;
+
+  void add(int fd) {
+    std::lock_guard<std::mutex> lock(mu);
+    fds.insert(fd);
+    active++;
+  }
+  void remove(int fd) {
+    std::lock_guard<std::mutex> lock(mu);
+    fds.erase(fd);
+    active--;
+    cv.notify_all();
+  }
+  void shutdown_all() {
+    std::lock_guard<std::mutex> lock(mu);
+    for (int fd : fds) ::shutdown(fd, SHUT_RDWR);
+  }
+  void wait_drained(int timeout_ms) {
+    std::unique_lock<std::mutex> lock(mu);
+    cv.wait_for(lock, std::chrono::milliseconds(timeout_ms), [&] { return active == 0; });
+  }
+};
+
 Server::Server(const std::string& bind_addr, uint16_t port, Handler handler)
-    : bind_addr_(bind_addr), port_(port), handler_(std::move(handler)) {}
+    : bind_addr_(bind_addr), port_(port), handler_(std::move(handler)) {
+  registry_ = std::make_shared<ConnRegistry>();
+}
 
 Server::~Server() { stop(); }
 
@@ -41,7 +76,7 @@ void Server::start() {
     ::getsockname(listen_fd_, reinterpret_cast<struct sockaddr*>(&addr), &len);
     port_ = ntohs(addr.sin_port);
   }
-  if (::listen(listen_fd_, 128) < 0) throw std::runtime_error("listen() failed");
+  if (::listen(listen_fd_, 256) < 0) throw std::runtime_error("listen() failed");
   running_.store(true);
   accept_thread_ = std::thread([this] { accept_loop(); });
 }
@@ -51,10 +86,8 @@ void Server::stop() {
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
   if (accept_thread_.joinable()) accept_thread_.join();
-  std::lock_guard<std::mutex> lock(workers_mu_);
-  for (auto& w : workers_)
-    if (w.joinable()) w.join();
-  workers_.clear();
+  registry_->shutdown_all();
+  registry_->wait_drained(5000);
 }
 
 void Server::accept_loop() {
@@ -66,14 +99,14 @@ void Server::accept_loop() {
     }
     int one = 1;
     ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
-    std::lock_guard<std::mutex> lock(workers_mu_);
-    // reap finished threads opportunistically
-    if (workers_.size() > 64) {
-      for (auto& w : workers_)
-        if (w.joinable()) w.join();
-      workers_.clear();
-    }
-    workers_.emplace_back([this, fd] { handle_conn(fd); });
+    registry_->add(fd);
+    // Detached per-connection handler; stop() tears sockets down via the
+    // registry instead of joining (a join would block on live keep-alives).
+    auto reg = registry_;
+    std::thread([this, fd, reg] {
+      handle_conn(fd);
+      reg->remove(fd);
+    }).detach();
   }
 }
 
@@ -145,6 +178,7 @@ void Server::handle_conn(int fd) {
     }
 
     const char* reason = resp.status == 200   ? "OK"
+                         : resp.status == 201 ? "Created"
                          : resp.status == 404 ? "Not Found"
                          : resp.status == 400 ? "Bad Request"
                                               : "Status";

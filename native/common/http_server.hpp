@@ -8,11 +8,15 @@
 #include <atomic>
 #include <functional>
 #include <map>
+#include <memory>
+#include <mutex>
 #include <string>
 #include <thread>
 #include <vector>
 
 namespace http {
+
+struct ConnRegistry;
 
 struct ServerRequest {
   std::string method;
@@ -50,8 +54,7 @@ private:
   int listen_fd_ = -1;
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
-  std::vector<std::thread> workers_;
-  std::mutex workers_mu_;
+  std::shared_ptr<ConnRegistry> registry_;
 };
 
 }  // namespace http

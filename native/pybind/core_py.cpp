@@ -17,6 +17,7 @@
 #include "../pruner/prom.hpp"
 #include "../pruner/promql.hpp"
 #include "../pruner/resources.hpp"
+#include "../pruner/synthbench.hpp"
 
 namespace py = pybind11;
 using namespace pruner;
@@ -214,19 +215,56 @@ PYBIND11_MODULE(_pruner_core, m) {
           auto prom = build_prom_client(cfg);
           KubeClient kube(KubeConfig::resolve());
           std::string query = build_idle_query(cfg.query_args());
-          QueryOutcome out = run_query_and_scale(*prom, kube, query, cfg, nullptr);
+          jsn::Value result = prom->query_vector(query);
+          QueryOutcome out;
+          std::vector<ScaleKind> roots = evaluate_candidates(kube, result, cfg, &out);
+          size_t scaled = 0;
+          if (cfg.run_mode == RunMode::ScaleDown) {
+            uint8_t enabled = get_enabled_resources(cfg.enabled_resources);
+            for (const auto& sk : roots) {
+              if (!(enabled & kind_flag(sk.kind))) continue;
+              scale(kube, sk);
+              scaled++;
+            }
+          }
           py::gil_scoped_acquire gil;
           py::dict d;
           d["num_series"] = out.num_series;
           d["num_unique_pods"] = out.num_unique_pods;
           d["shutdown_events"] = out.shutdown_events;
+          d["scaled"] = scaled;
           return d;
         },
         py::call_guard<py::gil_scoped_release>(),
-        "One full decision tick: Prometheus query + evaluation (dry-run unless "
-        "cfg.run_mode=scale-down, in which case scaling happens inline)");
+        "One full decision tick: Prometheus query + evaluation + (in "
+        "scale-down mode) inline actuation of every selected root");
 
   m.def("get_prometheus_token", [] { return get_prometheus_token(); });
+
+  // ---- native synthetic backend (benchmark harness) ----
+  py::class_<SyntheticBackend>(m, "SyntheticBackend")
+      .def(py::init([](int n_pods, int pods_per_parent, int gpus_per_pod, int latency_us,
+                       const std::string& model_name) {
+             SynthOptions o;
+             o.n_pods = n_pods;
+             o.pods_per_parent = pods_per_parent;
+             o.gpus_per_pod = gpus_per_pod;
+             o.latency_us = latency_us;
+             if (!model_name.empty()) o.model_name = model_name;
+             return new SyntheticBackend(o);
+           }),
+           py::arg("n_pods") = 1000, py::arg("pods_per_parent") = 2,
+           py::arg("gpus_per_pod") = 1, py::arg("latency_us") = 0,
+           py::arg("model_name") = "")
+      .def("start", &SyntheticBackend::start)
+      .def("stop", &SyntheticBackend::stop)
+      .def_property_readonly("prom_url", &SyntheticBackend::prom_url)
+      .def_property_readonly("k8s_url", &SyntheticBackend::k8s_url)
+      .def("set_series_value", &SyntheticBackend::set_series_value)
+      .def_property_readonly("events_posted", &SyntheticBackend::events_posted)
+      .def_property_readonly("scale_patches", &SyntheticBackend::scale_patches)
+      .def_property_readonly("requests_served", &SyntheticBackend::requests_served)
+      .def_property_readonly("expected_parents", &SyntheticBackend::expected_parents);
 
   m.def("counters_snapshot", [] {
     py::dict d;

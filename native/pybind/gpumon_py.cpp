@@ -1,0 +1,160 @@
+// gpumon_py.cpp — _gpumon: Python bindings over the mi355-exporter internals.
+//
+// Gives pytest and bench.py direct access to the ROCm sampler, the pod
+// attribution chain, and the metrics renderer — the same native objects the
+// mi355-exporter binary runs. GPU-marked tests exercise Sampler on real
+// gfx950; the attribution/rendering paths are CPU-testable via the
+// GPU_EXPORTER_*_ROOT / POD_MAP_FILE overrides.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "../common/json.hpp"
+#include "../common/log.hpp"
+#include "../exporter/attrib.hpp"
+#include "../exporter/registry.hpp"
+#include "../exporter/sampler.hpp"
+
+namespace py = pybind11;
+using namespace exporter;
+
+namespace {
+
+py::dict sample_to_dict(const DeviceSample& d) {
+  py::dict s;
+  s["index"] = d.index;
+  s["model_name"] = d.model_name;
+  s["unique_id"] = d.unique_id;
+  s["pci_bdf"] = d.pci_bdf;
+  s["drm_render_minor"] = d.drm_render_minor;
+  s["kfd_gpu_id"] = d.kfd_gpu_id;
+  s["busy_percent"] = d.busy_percent;
+  s["gr_engine_active"] = d.gr_engine_active;
+  s["mem_busy_percent"] = d.mem_busy_percent;
+  s["power_w"] = d.power_w;
+  s["vram_used_b"] = d.vram_used_b;
+  s["vram_total_b"] = d.vram_total_b;
+  s["temp_edge_c"] = d.temp_edge_c;
+  s["gfx_clock_mhz"] = d.gfx_clock_mhz;
+  s["energy_j"] = d.energy_j;
+  s["metrics_table_ok"] = d.metrics_table_ok;
+  s["gfx_activity_acc"] = d.gfx_activity_acc;
+  s["firmware_timestamp"] = d.firmware_timestamp;
+  return s;
+}
+
+DeviceSample sample_from_json(const jsn::Value& v) {
+  DeviceSample d;
+  d.index = static_cast<uint32_t>(v.get("index").as_int(0));
+  d.model_name = v.get("model_name").as_string_or("AMD Instinct MI355X");
+  d.unique_id = v.get("unique_id").as_string_or("");
+  d.drm_render_minor = static_cast<uint32_t>(v.get("drm_render_minor").as_int(128));
+  d.kfd_gpu_id = static_cast<uint64_t>(v.get("kfd_gpu_id").as_int(0));
+  d.busy_percent = v.get("busy_percent").as_double(0);
+  d.gr_engine_active = v.get("gr_engine_active").as_double(0);
+  d.mem_busy_percent = v.get("mem_busy_percent").as_double(0);
+  d.power_w = v.get("power_w").as_double(0);
+  d.vram_used_b = v.get("vram_used_b").as_double(0);
+  d.vram_total_b = v.get("vram_total_b").as_double(0);
+  d.temp_edge_c = v.get("temp_edge_c").as_double(0);
+  d.gfx_clock_mhz = v.get("gfx_clock_mhz").as_double(0);
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_gpumon, m) {
+  m.doc() = "native mi355-exporter internals: ROCm sampler, attribution, registry";
+
+  logx::init(logx::Format::Default);
+
+  py::register_exception<SamplerError>(m, "SamplerError");
+
+  py::class_<Sampler>(m, "Sampler")
+      .def(py::init<int>(), py::arg("poll_interval_ms") = 1000)
+      .def("init", &Sampler::init,
+           "Initialize rocm_smi and enumerate devices (raises SamplerError "
+           "without an AMD GPU)")
+      .def("start", &Sampler::start)
+      .def("stop", &Sampler::stop)
+      .def("poll_once", &Sampler::poll_once, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("device_count", &Sampler::device_count)
+      .def("snapshot",
+           [](Sampler& s, bool reset_window) {
+             auto samples = s.snapshot(reset_window);
+             py::list out;
+             for (const auto& d : samples) out.append(sample_to_dict(d));
+             return out;
+           },
+           py::arg("reset_window") = false);
+
+  m.def("pod_uid_from_cgroup", [](const std::string& text) -> py::object {
+    auto uid = pod_uid_from_cgroup(text);
+    return uid ? py::cast(*uid) : py::none();
+  });
+
+  m.def("kfd_gpu_pids", [] {
+    py::dict out;
+    for (const auto& [gpu_id, pids] : kfd_gpu_pids()) out[py::cast(gpu_id)] = pids;
+    return out;
+  });
+
+  py::class_<Attributor>(m, "Attributor")
+      .def(py::init<>())
+      .def("resolve",
+           [](Attributor& a, const std::vector<std::pair<uint32_t, uint64_t>>& idx_kfd) {
+             // release the GIL during the native work: the apiserver this
+             // talks to may be an in-process Python fixture
+             std::map<uint32_t, PodAttribution> resolved;
+             {
+               py::gil_scoped_release nogil;
+               resolved = a.resolve(idx_kfd);
+             }
+             py::dict out;
+             for (const auto& [idx, attr] : resolved) {
+               py::dict v;
+               v["pod"] = attr.pod;
+               v["namespace"] = attr.ns;
+               v["container"] = attr.container;
+               out[py::cast(idx)] = v;
+             }
+             return out;
+           })
+      .def("lookup_uid", [](Attributor& a, const std::string& uid) -> py::object {
+        std::optional<PodAttribution> attr;
+        {
+          py::gil_scoped_release nogil;
+          attr = a.lookup_uid(uid);
+        }
+        if (!attr) return py::none();
+        py::dict v;
+        v["pod"] = attr->pod;
+        v["namespace"] = attr->ns;
+        v["container"] = attr->container;
+        return v;
+      });
+
+  // Render synthetic samples (JSON list) — lets the exposition format be
+  // pinned by CPU tests.
+  m.def("render_metrics",
+        [](const std::string& samples_json, const std::string& attribs_json,
+           const std::string& hostname, const std::string& node_type) {
+          jsn::Value sv = jsn::parse(samples_json);
+          std::vector<DeviceSample> samples;
+          for (const auto& v : sv.arr()) samples.push_back(sample_from_json(v));
+          std::map<uint32_t, PodAttribution> attribs;
+          if (!attribs_json.empty()) {
+            jsn::Value av = jsn::parse(attribs_json);
+            for (const auto& [k, v] : av.obj()) {
+              attribs[static_cast<uint32_t>(std::stoul(k))] = PodAttribution{
+                  v.get("pod").as_string(), v.get("namespace").as_string(),
+                  v.get("container").as_string_or("")};
+            }
+          }
+          RenderOptions opts;
+          opts.hostname = hostname;
+          if (!node_type.empty()) opts.const_labels.emplace_back("node_type", node_type);
+          return render_metrics(samples, attribs, opts);
+        },
+        py::arg("samples_json"), py::arg("attribs_json") = "",
+        py::arg("hostname") = "test-node", py::arg("node_type") = "");
+}

@@ -1,0 +1,235 @@
+"""GPU tests — require a real MI355X (gfx950); run via gpurun / the driver.
+
+Prove the counter-semantics contract on silicon (SURVEY.md §7 "Hard parts"):
+the first-party sampler must read exactly 0 on an idle GPU and rise under the
+gfx950 HIP busy probe, in both the instantaneous busy percent and the
+windowed GR_ENGINE_ACTIVE ratio — then the whole culler stack must make the
+right decision from that real signal.
+"""
+
+import json
+import os
+import subprocess
+import time
+import urllib.request
+from pathlib import Path
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+REPO_ROOT = Path(__file__).resolve().parent.parent
+
+
+def _require_gpu():
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU visible")
+
+
+@pytest.fixture(scope="module")
+def sampler():
+    _require_gpu()
+    from gpu_pruner_amd import _gpumon
+
+    s = _gpumon.Sampler(poll_interval_ms=100)
+    s.init()  # must not silently fall back — raises SamplerError if broken
+    yield s
+    s.stop()
+
+
+def _settle_idle(sampler, seconds=5.0, target=0.0):
+    """Poll until busy_percent settles at `target` (returns last value)."""
+    deadline = time.monotonic() + seconds
+    busy = None
+    while time.monotonic() < deadline:
+        sampler.poll_once()
+        busy = sampler.snapshot()[0]["busy_percent"]
+        if busy == target:
+            return busy
+        time.sleep(0.1)
+    return busy
+
+
+def test_sampler_enumerates_mi355x(sampler):
+    assert sampler.device_count >= 1
+    snap = sampler.snapshot()[0]
+    assert snap["model_name"], "model name must be non-empty"
+    # 288 GB HBM3E per MI355X
+    assert snap["vram_total_b"] > 200 * 2**30, snap["vram_total_b"]
+    assert snap["pci_bdf"], "PCI BDF must resolve"
+
+
+def test_idle_utilization_is_exactly_zero(sampler):
+    """The PromQL `== 0` predicate depends on a literal zero when idle."""
+    busy = _settle_idle(sampler, seconds=10.0)
+    assert busy == 0.0, f"idle GPU reports busy_percent={busy}"
+    # windowed ratio over a fresh idle window
+    sampler.snapshot(True)  # reset window
+    time.sleep(1.0)
+    sampler.poll_once()
+    ratio = sampler.snapshot()[0]["gr_engine_active"]
+    assert ratio < 0.01, f"idle GR_ENGINE_ACTIVE={ratio}"
+
+
+def test_busy_probe_raises_utilization(sampler):
+    from gpu_pruner_amd import probe
+
+    _settle_idle(sampler, seconds=5.0)
+    sampler.snapshot(True)
+    probe.start(device=0, max_seconds=30.0)
+    try:
+        busy = 0.0
+        for _ in range(40):
+            time.sleep(0.1)
+            sampler.poll_once()
+            busy = sampler.snapshot()[0]["busy_percent"]
+            if busy >= 90.0:
+                break
+    finally:
+        probe.stop()
+    assert busy >= 90.0, f"probe should saturate the GPU (busy={busy})"
+    ratio = sampler.snapshot()[0]["gr_engine_active"]
+    assert ratio > 0.3, f"windowed ratio under load = {ratio}"
+    # and utilization must fall back to zero afterwards
+    busy = _settle_idle(sampler, seconds=10.0)
+    assert busy == 0.0, f"busy stuck at {busy} after probe stop"
+
+
+def test_power_and_clock_sane(sampler):
+    sampler.poll_once()
+    snap = sampler.snapshot()[0]
+    assert 20.0 < snap["power_w"] < 1600.0, snap["power_w"]  # idle..TDP range
+    assert snap["metrics_table_ok"], "gpu_metrics table must be readable"
+
+
+def test_activity_acc_counter_advances_under_load(sampler):
+    """The firmware gfx_activity_acc accumulator must advance under load —
+    it is the GRBM_GUI_ACTIVE-derived signal for the windowed ratio."""
+    from gpu_pruner_amd import probe
+
+    sampler.poll_once()
+    before = sampler.snapshot()[0]["gfx_activity_acc"]
+    with probe.busy_load(device=0, max_seconds=15.0):
+        time.sleep(1.5)
+        sampler.poll_once()
+    after = sampler.snapshot()[0]["gfx_activity_acc"]
+    assert after > before, f"gfx_activity_acc did not advance ({before} -> {after})"
+
+
+def test_exporter_binary_serves_real_metrics():
+    _require_gpu()
+    port = 19400
+    proc = subprocess.Popen(
+        [str(REPO_ROOT / "bin" / "mi355-exporter"), "-p", str(port),
+         "-b", "127.0.0.1", "-i", "200", "--node-type", "amd-mi355x"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        text = None
+        for _ in range(50):
+            time.sleep(0.2)
+            if proc.poll() is not None:
+                raise AssertionError(
+                    f"exporter died: {proc.stderr.read().decode()[:500]}")
+            try:
+                text = urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
+                break
+            except OSError:
+                continue
+        assert text is not None, "exporter never served /metrics"
+        assert "DCGM_FI_PROF_GR_ENGINE_ACTIVE{" in text
+        assert "DCGM_FI_DEV_GPU_UTIL{" in text
+        assert "DCGM_FI_DEV_POWER_USAGE{" in text
+        assert 'node_type="amd-mi355x"' in text
+        assert 'modelName=' in text
+        health = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/healthz", timeout=2).read()
+        assert health == b"ok\n"
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
+
+
+def _parse_prom_text(text, family):
+    """family{labels} value → list of (labels_dict, float)."""
+    out = []
+    for line in text.splitlines():
+        if not line.startswith(family + "{"):
+            continue
+        labels_part, value = line.rsplit("} ", 1)
+        labels = {}
+        for item in labels_part[len(family) + 1:].split('",'):
+            if "=" in item:
+                k, v = item.split("=", 1)
+                labels[k.strip(",")] = v.strip('"')
+        out.append((labels, float(value)))
+    return out
+
+
+def test_full_stack_idle_cull_from_real_gpu_signal(fake_api, fake_prom, pruner_bin):
+    """The headline e2e (BASELINE config 2 analog): a real idle MI355X, its
+    activity scraped from the live mi355-exporter, drives the culler to scale
+    an idle Notebook to zero."""
+    _require_gpu()
+    port = 19401
+    exporter = subprocess.Popen(
+        [str(REPO_ROOT / "bin" / "mi355-exporter"), "-p", str(port),
+         "-b", "127.0.0.1", "-i", "200"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        text = None
+        for _ in range(50):
+            time.sleep(0.2)
+            try:
+                text = urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
+                break
+            except OSError:
+                continue
+        assert text, "exporter unreachable"
+        time.sleep(1.0)  # let an idle window accumulate
+        text = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
+        series = _parse_prom_text(text, "DCGM_FI_PROF_GR_ENGINE_ACTIVE")
+        assert series, "no activity series"
+        gpu0 = [s for s in series if s[0].get("gpu") == "0"][0]
+        activity = gpu0[1]
+        assert activity == 0.0, f"idle GPU activity={activity}"
+
+        # stand in for Prometheus: the scraped value becomes the query result
+        nb = fake_api.add_notebook("workbench", "ml")
+        fake_api.add_statefulset("workbench-ss", "ml", notebook_owner=nb)
+        fake_api.add_pod("workbench-ss-0", "ml", owner_kind="StatefulSet",
+                         owner_name="workbench-ss", age_s=3 * 3600)
+        fake_prom.add_idle_series("workbench-ss-0", "ml", value=activity,
+                                  model_name=gpu0[0].get("modelName", "AMD"))
+
+        env = dict(os.environ)
+        env["GPU_PRUNER_K8S_URL"] = fake_api.url
+        env["PROMETHEUS_TOKEN"] = "t"
+        r = subprocess.run(
+            [pruner_bin, "--prometheus-url", fake_prom.url, "--run-mode", "scale-down"],
+            capture_output=True, text=True, timeout=60, env=env)
+        assert r.returncode == 0, r.stderr
+        nb_after = fake_api.get("Notebook", "ml", "workbench")
+        assert "kubeflow-resource-stopped" in nb_after["metadata"].get("annotations", {})
+    finally:
+        exporter.terminate()
+        exporter.wait(timeout=10)
+
+
+def test_bench_one_gpu_quick():
+    """bench.py runs with the real sampler and prints the contract line."""
+    _require_gpu()
+    r = subprocess.run(
+        [os.environ.get("PYTHON", "python3"), str(REPO_ROOT / "bench.py"),
+         "--steps", "5", "--warmup", "1", "--pods", "200"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO_ROOT))
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    result = json.loads(line)
+    assert result["metric"] == "pods_evaluated_per_sec"
+    assert result["value"] > 0
+    assert "real GPU" in result["config"]["utilization_source"]
