@@ -1,0 +1,93 @@
+"""FakeOtlpCollector — minimal OTLP/HTTP+JSON collector double.
+
+Records every payload POSTed to /v1/traces and /v1/metrics so tests can
+assert the daemon's span + counter export surface.
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+
+class FakeOtlpCollector:
+    def __init__(self, host: str = "127.0.0.1", port: int = 0):
+        self._lock = threading.Lock()
+        self.traces: list[dict] = []
+        self.metrics: list[dict] = []
+
+        fixture = self
+
+        class Handler(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+            disable_nagle_algorithm = True
+
+            def log_message(self, *args):
+                pass
+
+            def do_POST(self):
+                length = int(self.headers.get("Content-Length", "0"))
+                try:
+                    payload = json.loads(self.rfile.read(length) or b"{}")
+                except json.JSONDecodeError:
+                    payload = {}
+                with fixture._lock:
+                    if self.path.endswith("/v1/traces"):
+                        fixture.traces.append(payload)
+                    elif self.path.endswith("/v1/metrics"):
+                        fixture.metrics.append(payload)
+                body = b"{}"
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+        self._server = ThreadingHTTPServer((host, port), Handler)
+        self._server.daemon_threads = True
+        self._thread = threading.Thread(
+            target=lambda: self._server.serve_forever(poll_interval=0.05), daemon=True)
+
+    def start(self):
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._server.shutdown()
+        self._server.server_close()
+
+    @property
+    def url(self) -> str:
+        host, port = self._server.server_address[:2]
+        return f"http://{host}:{port}"
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()
+
+    # -- helpers --------------------------------------------------------------
+    def span_names(self) -> list[str]:
+        with self._lock:
+            names = []
+            for payload in self.traces:
+                for rs in payload.get("resourceSpans", []):
+                    for ss in rs.get("scopeSpans", []):
+                        names.extend(s.get("name") for s in ss.get("spans", []))
+            return names
+
+    def metric_points(self) -> dict[str, int]:
+        """latest value per metric name across all exports."""
+        with self._lock:
+            out = {}
+            for payload in self.metrics:
+                for rm in payload.get("resourceMetrics", []):
+                    for sm in rm.get("scopeMetrics", []):
+                        for m in sm.get("metrics", []):
+                            dps = (m.get("sum") or m.get("gauge") or {}).get(
+                                "dataPoints", [])
+                            if dps:
+                                out[m["name"]] = int(dps[-1].get("asInt", 0))
+            return out

@@ -15,7 +15,8 @@ from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 
 
 class FakePrometheus:
-    def __init__(self, host: str = "127.0.0.1", port: int = 0):
+    def __init__(self, host: str = "127.0.0.1", port: int = 0,
+                 certfile: str | None = None, keyfile: str | None = None):
         self._lock = threading.Lock()
         self.series: list[dict] = []
         self.queries: list[str] = []
@@ -79,6 +80,13 @@ class FakePrometheus:
 
         self._server = ThreadingHTTPServer((host, port), Handler)
         self._server.daemon_threads = True
+        self._tls = certfile is not None
+        if certfile is not None:
+            import ssl
+
+            ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            ctx.load_cert_chain(certfile, keyfile)
+            self._server.socket = ctx.wrap_socket(self._server.socket, server_side=True)
         self._thread = threading.Thread(
             target=lambda: self._server.serve_forever(poll_interval=0.05), daemon=True)
 
@@ -94,7 +102,8 @@ class FakePrometheus:
     @property
     def url(self) -> str:
         host, port = self._server.server_address[:2]
-        return f"http://{host}:{port}"
+        scheme = "https" if self._tls else "http"
+        return f"{scheme}://{host}:{port}"
 
     def __enter__(self):
         return self.start()

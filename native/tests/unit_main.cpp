@@ -1,0 +1,139 @@
+// unit_main.cpp — native unit tests for the sanitizer CI tier.
+//
+// The Python suite covers behavior; this binary re-exercises the pure-logic
+// core (JSON, query builder, scaling model, queue, CLI) so it can be built
+// under -fsanitize=address,undefined / thread and run with zero Python in
+// the loop (the Rust reference gets safety from the
+// compiler + clippy; the C++ build gets ASan/TSan CI jobs as the
+// equivalent).
+#include <cassert>
+#include <cstdio>
+#include <thread>
+#include <vector>
+
+#include "../common/json.hpp"
+#include "../common/queue.hpp"
+#include "../common/strutil.hpp"
+#include "../pruner/config.hpp"
+#include "../pruner/promql.hpp"
+#include "../pruner/resources.hpp"
+
+using namespace pruner;
+
+static int failures = 0;
+#define CHECK(cond)                                                   \
+  do {                                                                \
+    if (!(cond)) {                                                    \
+      std::fprintf(stderr, "FAIL %s:%d: %s\n", __FILE__, __LINE__, #cond); \
+      failures++;                                                     \
+    }                                                                 \
+  } while (0)
+
+static void test_json() {
+  jsn::Value v = jsn::parse(R"({"a": [1, 2.5, "x", null, true], "b": {"c": "dé"}})");
+  CHECK(v.get("a").size() == 5);
+  CHECK(v.at({"b", "c"}).as_string() == "d\xc3\xa9");
+  CHECK(jsn::parse(v.dump()) == v);
+  jsn::Value obj = jsn::parse(R"({"spec": {"replicas": 3, "keep": 1}})");
+  obj.merge_patch(jsn::parse(R"({"spec": {"replicas": 0}})"));
+  CHECK(obj.at({"spec", "replicas"}).as_int() == 0);
+  CHECK(obj.at({"spec", "keep"}).as_int() == 1);
+  obj.merge_patch(jsn::parse(R"({"spec": {"keep": null}})"));
+  CHECK(!obj.get("spec").contains("keep"));
+  bool threw = false;
+  try {
+    jsn::parse("{broken");
+  } catch (const jsn::ParseError&) {
+    threw = true;
+  }
+  CHECK(threw);
+}
+
+static void test_promql() {
+  QueryArgs qa;
+  qa.duration_min = 45;
+  std::string q = build_idle_query(qa);
+  CHECK(q.find("max_over_time(") != std::string::npos);
+  CHECK(q.find("avg_over_time") == std::string::npos);
+  CHECK(q.find("[45m]") != std::string::npos);
+  CHECK(q.find("unless") == std::string::npos);
+  qa.power_threshold_w = 150.0;
+  qa.namespace_re = "ml-team";
+  q = build_idle_query(qa);
+  CHECK(q.find("unless on (exported_pod, exported_namespace)") != std::string::npos);
+  CHECK(q.find(">= 150") != std::string::npos);
+  CHECK(strutil::count_occurrences(q, "exported_namespace =~ \"ml-team\"") == 5);
+}
+
+static void test_resources() {
+  CHECK(get_enabled_resources("drsin") == RK_ALL);
+  CHECK(get_enabled_resources("xdqz") == RK_DEPLOYMENT);
+  CHECK(get_enabled_resources("") == RK_NONE);
+  ScaleKind a{Kind::Notebook, jsn::parse(R"({"metadata":{"name":"a","namespace":"ns","uid":"u1"}})")};
+  ScaleKind b{Kind::Notebook, jsn::parse(R"({"metadata":{"name":"b","namespace":"ns","uid":"u1"}})")};
+  CHECK(a == b);  // CRDs: uid equality
+  ScaleKindSet set;
+  set.insert(a);
+  set.insert(b);
+  CHECK(set.size() == 1);
+  jsn::Value ev = generate_scale_event(a);
+  CHECK(ev.get("reason").as_string() == "Pod ns::a was not using GPU");
+  CHECK(strutil::starts_with(ev.at({"metadata", "name"}).as_string(), "gpuscaler-"));
+  CHECK(object_path(Kind::InferenceService, "ns", "x") ==
+        "/apis/serving.kserve.io/v1beta1/namespaces/ns/inferenceservices/x");
+}
+
+static void test_queue_mpmc() {
+  // TSan target: hammer the bounded queue from several producers/consumers.
+  qx::BoundedQueue<int> q(16);
+  std::atomic<long> sum{0};
+  std::vector<std::thread> threads;
+  for (int p = 0; p < 4; p++)
+    threads.emplace_back([&q, p] {
+      for (int i = 0; i < 1000; i++) q.push(p * 1000 + i);
+    });
+  for (int c = 0; c < 4; c++)
+    threads.emplace_back([&q, &sum] {
+      while (auto v = q.pop()) sum.fetch_add(*v);
+    });
+  for (int p = 0; p < 4; p++) threads[static_cast<size_t>(p)].join();
+  q.close();
+  for (size_t c = 4; c < threads.size(); c++) threads[c].join();
+  long expect = 0;
+  for (int p = 0; p < 4; p++)
+    for (int i = 0; i < 1000; i++) expect += p * 1000 + i;
+  CHECK(sum.load() == expect);
+}
+
+static void test_cli() {
+  auto r = parse_cli({"--prometheus-url", "http://p", "-t", "15", "--run-mode",
+                      "scale-down"});
+  CHECK(!r.error);
+  CHECK(r.config.duration_min == 15);
+  CHECK(r.config.run_mode == RunMode::ScaleDown);
+  CHECK(parse_cli({}).error.has_value());
+  CHECK(parse_cli({"--prometheus-url", "http://p", "--bogus"}).error.has_value());
+}
+
+static void test_strutil() {
+  double ts = 0;
+  CHECK(strutil::parse_rfc3339("2026-01-02T03:04:05Z", &ts));
+  CHECK(ts == 1767323045.0);  // date -u -d 2026-01-02T03:04:05Z +%s
+  double ts2 = 0;
+  CHECK(strutil::parse_rfc3339("2026-01-02T04:04:05+01:00", &ts2));
+  CHECK(ts == ts2);  // same instant
+  CHECK(!strutil::parse_rfc3339("not a date", &ts));
+  CHECK(strutil::uuid4_simple().size() == 32);
+  CHECK(strutil::uuid4_simple() != strutil::uuid4_simple());
+}
+
+int main() {
+  test_json();
+  test_promql();
+  test_resources();
+  test_queue_mpmc();
+  test_cli();
+  test_strutil();
+  if (failures == 0) std::puts("native unit tests: all passed");
+  return failures == 0 ? 0 : 1;
+}

@@ -1,0 +1,101 @@
+"""Observability tests: OTLP export, log formats, counters.
+
+The reference exports spans + tracing-derived counters over OTLP behind the
+`otel` feature, env-configured (reference
+main.rs:138-271). This build speaks OTLP/HTTP+JSON (native/pruner/otlp.cpp);
+these tests run the real daemon binary against a fake collector and pin the
+span + counter surface, plus the three log formats.
+"""
+
+import json
+import os
+import subprocess
+
+import pytest
+
+from gpu_pruner_amd.fixtures import FakeOtlpCollector
+
+
+def run_daemon(pruner_bin, fake_api, fake_prom, *args, env_extra=None, timeout=60):
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    env.pop("OTEL_EXPORTER_OTLP_ENDPOINT", None)
+    if env_extra:
+        env.update(env_extra)
+    return subprocess.run(
+        [pruner_bin, "--prometheus-url", fake_prom.url, *args],
+        capture_output=True, text=True, timeout=timeout, env=env)
+
+
+@pytest.fixture
+def idle_cluster(fake_api, fake_prom):
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    fake_api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p0", "ml")
+    return fake_api
+
+
+def test_otlp_spans_and_counters_exported(pruner_bin, idle_cluster, fake_prom):
+    with FakeOtlpCollector() as collector:
+        r = run_daemon(pruner_bin, idle_cluster, fake_prom, "--run-mode", "scale-down",
+                       env_extra={"OTEL_EXPORTER_OTLP_ENDPOINT": collector.url,
+                                  "OTEL_METRIC_EXPORT_INTERVAL": "60000"})
+        assert r.returncode == 0, r.stderr
+        # shutdown() final-flushes, so one-shot runs still export
+        assert "run_query_and_scale" in collector.span_names()
+        points = collector.metric_points()
+        assert points.get("query_successes") == 1
+        assert points.get("scale_successes") == 1
+        assert points.get("query_returned_candidates") == 1
+        assert points.get("query_returned_shutdown_events") == 1
+
+
+def test_otlp_failure_counters(pruner_bin, fake_api, fake_prom):
+    fake_prom.fail_next = 100
+    with FakeOtlpCollector() as collector:
+        r = run_daemon(pruner_bin, fake_api, fake_prom, "--daemon-mode",
+                       "--check-interval", "0", "--max-failures", "1",
+                       env_extra={"OTEL_EXPORTER_OTLP_ENDPOINT": collector.url,
+                                  "OTEL_METRIC_EXPORT_INTERVAL": "60000"})
+        assert r.returncode != 0
+        points = collector.metric_points()
+        assert points.get("query_failures", 0) >= 2
+
+
+def test_otlp_disabled_without_endpoint(pruner_bin, idle_cluster, fake_prom):
+    with FakeOtlpCollector() as collector:
+        run_daemon(pruner_bin, idle_cluster, fake_prom,
+                   env_extra={"OTEL_SDK_DISABLED": "true",
+                              "OTEL_EXPORTER_OTLP_ENDPOINT": collector.url})
+        assert collector.traces == []
+        assert collector.metrics == []
+
+
+def test_log_format_json(pruner_bin, idle_cluster, fake_prom):
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom, "--log-format", "json")
+    assert r.returncode == 0
+    json_lines = [l for l in r.stderr.splitlines() if l.startswith("{")]
+    assert json_lines, "json format should emit JSON log lines"
+    rec = json.loads(json_lines[0])
+    assert {"timestamp", "level", "target", "fields"} <= set(rec)
+    assert "message" in rec["fields"]
+
+
+def test_log_format_default(pruner_bin, idle_cluster, fake_prom):
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom, "--log-format", "default")
+    assert r.returncode == 0
+    assert any(" INFO " in l for l in r.stderr.splitlines())
+
+
+def test_log_level_filter_env(pruner_bin, idle_cluster, fake_prom):
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom,
+                   env_extra={"GPU_PRUNER_LOG": "error"})
+    assert r.returncode == 0
+    assert not any(" INFO " in l for l in r.stderr.splitlines())
+    # RUST_LOG is honored too (drop-in with the reference's env knob)
+    r2 = run_daemon(pruner_bin, idle_cluster, fake_prom,
+                    env_extra={"RUST_LOG": "error", "GPU_PRUNER_LOG": ""})
+    assert not any(" INFO " in l for l in r2.stderr.splitlines())
