@@ -54,7 +54,7 @@ struct ClientOptions {
   std::string ca_file;         // PEM bundle for TlsVerify::CustomCa
   int connect_timeout_ms = 5000;
   int io_timeout_ms = 30000;   // per-request read/write deadline
-  int max_pool_per_origin = 64;
+  int max_pool_per_origin = 256;
 };
 
 class Connection;  // opaque: one TCP/TLS (or unix) stream

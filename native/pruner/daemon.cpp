@@ -2,7 +2,10 @@
 
 #include <atomic>
 #include <chrono>
+#include <memory>
+#include <mutex>
 #include <thread>
+#include <vector>
 
 #include "../common/log.hpp"
 #include "../common/queue.hpp"
@@ -34,18 +37,25 @@ int run_daemon(const Config& cfg) {
   qx::BoundedQueue<ScaleKind> queue(static_cast<size_t>(cfg.queue_capacity));
   std::atomic<int> exit_code{0};
 
-  std::thread consumer([&] {
-    // The consumer owns its own apiserver client (the reference builds a
-    // second KubeClient in scale_down_task, main.rs:333).
-    std::unique_ptr<KubeClient> kube;
+  // Consumer pool draining the scale queue. The reference uses one serial
+  // consumer (main.rs:332-367); each scale is 2 apiserver round-trips, so a
+  // small pool keeps actuation off the critical path at 1000-pod scale. The
+  // shared KubeClient's connection pool is reused across workers.
+  std::shared_ptr<KubeClient> consumer_kube;
+  std::mutex consumer_kube_mu;
+  auto consume = [&] {
     while (auto sk = queue.pop()) {
       if (!(enabled & kind_flag(sk->kind))) {
         LOGI(TARGET, "Skipping resource type " + sk->kind_str() + " because it is not enabled");
         continue;
       }
       try {
-        if (!kube) kube = std::make_unique<KubeClient>(KubeConfig::resolve());
-        scale(*kube, *sk);
+        {
+          std::lock_guard<std::mutex> lock(consumer_kube_mu);
+          if (!consumer_kube)
+            consumer_kube = std::make_shared<KubeClient>(KubeConfig::resolve());
+        }
+        scale(*consumer_kube, *sk);
       } catch (const std::exception& e) {
         logx::counter_add("monotonic_counter.scale_failures", 1);
         LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
@@ -55,7 +65,11 @@ int run_daemon(const Config& cfg) {
       LOGI(TARGET, "Scaled Resource: [" + sk->kind_str() + "] - " +
                        sk->ns().value_or("default") + ":" + sk->name());
     }
-  });
+  };
+  int n_consumers = std::max(1, std::min(cfg.max_concurrency, 8));
+  std::vector<std::thread> consumers;
+  consumers.reserve(static_cast<size_t>(n_consumers));
+  for (int i = 0; i < n_consumers; i++) consumers.emplace_back(consume);
 
   int consecutive_failures = 0;
   auto next_tick = std::chrono::steady_clock::now();
@@ -92,8 +106,8 @@ int run_daemon(const Config& cfg) {
     if (!cfg.daemon_mode) break;
   }
 
-  queue.close();  // producer done: consumer drains and exits
-  consumer.join();
+  queue.close();  // producer done: consumers drain and exit
+  for (auto& c : consumers) c.join();
   return exit_code.load();
 }
 

@@ -130,6 +130,42 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
   }
 }
 
+size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
+                 uint8_t enabled_mask, int concurrency) {
+  std::atomic<size_t> scaled{0};
+  std::atomic<size_t> next{0};
+  auto worker = [&]() {
+    while (true) {
+      size_t i = next.fetch_add(1, std::memory_order_relaxed);
+      if (i >= roots.size()) break;
+      const ScaleKind& sk = roots[i];
+      if (!(enabled_mask & kind_flag(sk.kind))) {
+        LOGI(TARGET, "Skipping resource type " + sk.kind_str() + " because it is not enabled");
+        continue;
+      }
+      try {
+        scale(kube, sk);
+        scaled.fetch_add(1, std::memory_order_relaxed);
+        logx::counter_add("monotonic_counter.scale_successes", 1);
+      } catch (const std::exception& e) {
+        logx::counter_add("monotonic_counter.scale_failures", 1);
+        LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
+      }
+    }
+  };
+  size_t n_workers = std::min<size_t>(static_cast<size_t>(std::max(concurrency, 1)),
+                                      roots.size());
+  if (n_workers <= 1) {
+    worker();
+  } else {
+    std::vector<std::thread> threads;
+    threads.reserve(n_workers);
+    for (size_t t = 0; t < n_workers; t++) threads.emplace_back(worker);
+    for (auto& t : threads) t.join();
+  }
+  return scaled.load();
+}
+
 std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& result_vector,
                                            const Config& cfg, QueryOutcome* outcome) {
   // Dedup series by (pod, namespace): multi-GPU pods emit one series per GPU

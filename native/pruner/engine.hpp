@@ -39,6 +39,14 @@ std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& po
 // scale-to-zero patch. Throws KubeError on patch failure.
 void scale(KubeClient& kube, const ScaleKind& sk);
 
+// Actuate a batch of roots concurrently (each root = Event POST + patch, 2
+// apiserver round-trips): skips kinds not in `enabled_mask`, counts
+// successes, logs failures. The reference drains its scale channel with a
+// single serial consumer (main.rs:332-367); with per-root RTTs this is the
+// second fan-out that matters at 1000-pod scale.
+size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
+                 uint8_t enabled_mask, int concurrency);
+
 struct QueryOutcome {
   size_t num_series = 0;       // raw series returned by Prometheus
   size_t num_unique_pods = 0;  // after (pod, namespace) dedup

@@ -221,11 +221,7 @@ PYBIND11_MODULE(_pruner_core, m) {
           size_t scaled = 0;
           if (cfg.run_mode == RunMode::ScaleDown) {
             uint8_t enabled = get_enabled_resources(cfg.enabled_resources);
-            for (const auto& sk : roots) {
-              if (!(enabled & kind_flag(sk.kind))) continue;
-              scale(kube, sk);
-              scaled++;
-            }
+            scaled = scale_all(kube, roots, enabled, cfg.max_concurrency);
           }
           py::gil_scoped_acquire gil;
           py::dict d;
