@@ -193,16 +193,21 @@ void Sampler::poll_device(uint32_t i) {
     if (dt > 0) {
       double ratio = d.busy_percent / 100.0;
       // When the firmware accumulator advanced, derive the exact ratio over
-      // the poll interval from Δacc/Δt: gfx_activity_acc accumulates
-      // percent·time ticks on the 100 Hz firmware clock, so the windowed
-      // ratio is Δacc / (Δfw_ts_in_s × 100) with fw ts in 10 ns units.
+      // the poll interval from Δacc/Δt. Units calibrated on MI355X silicon
+      // (profiles/raw/calibration.log): firmware_timestamp ticks in ns on
+      // gfx950 (the rocm_smi header documents 10 ns — we auto-detect by
+      // comparing against the host monotonic delta), and gfx_activity_acc
+      // accumulates 100,000 counts per second at 100% busy.
       if (d.metrics_table_ok && st.have_prev_acc &&
           d.firmware_timestamp > st.prev_fw_ts) {
+        double fw_dt = static_cast<double>(d.firmware_timestamp - st.prev_fw_ts);
+        double fw_dt_ns = fw_dt * 1e-9, fw_dt_10ns = fw_dt * 1e-8;
         double fw_dt_s =
-            static_cast<double>(d.firmware_timestamp - st.prev_fw_ts) * 1e-8;
+            std::abs(fw_dt_ns - dt) <= std::abs(fw_dt_10ns - dt) ? fw_dt_ns : fw_dt_10ns;
+        constexpr double kAccFullRate = 100000.0;  // counts/s at 100% (measured)
         if (fw_dt_s > 0) {
-          double acc_ratio =
-              static_cast<double>(d.gfx_activity_acc - st.prev_acc) / (fw_dt_s * 100.0);
+          double acc_ratio = static_cast<double>(d.gfx_activity_acc - st.prev_acc) /
+                             (fw_dt_s * kAccFullRate);
           if (acc_ratio >= 0.0 && acc_ratio <= 1.5) ratio = std::min(acc_ratio, 1.0);
         }
       }

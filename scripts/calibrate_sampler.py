@@ -51,10 +51,13 @@ def main():
     d = s.snapshot()[0]
     if acc0:
         dacc = d["gfx_activity_acc"] - acc0[0]
-        dfw_s = (d["firmware_timestamp"] - acc0[1]) * 1e-8
+        dfw = d["firmware_timestamp"] - acc0[1]
+        # gfx950 firmware_timestamp ticks in ns (header says 10 ns; verified
+        # against wall time in profiles/raw/calibration.log)
+        dfw_s = dfw * 1e-9
         rate = dacc / max(dfw_s, 1e-9)
-        print(f"calib: dacc={dacc} dfw_s={dfw_s:.3f} rate={rate:.1f}/s "
-              f"(sampler model expects ~100/s at 100% busy)")
+        print(f"calib: dacc={dacc} dfw_ns={dfw} rate={rate:.1f}/s at 100% busy "
+              f"(sampler model uses 100000/s)")
     print("--- settle ---")
     for _ in range(10):
         time.sleep(0.5)
