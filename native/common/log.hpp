@@ -38,11 +38,18 @@ void emit(Level lvl, const std::string& target, const std::string& msg);
 void emit_kv(Level lvl, const std::string& target, const std::string& msg,
              const std::vector<std::pair<std::string, std::string>>& fields);
 
-#define LOGT(target, msg) ::logx::emit(::logx::Level::Trace, target, msg)
-#define LOGD(target, msg) ::logx::emit(::logx::Level::Debug, target, msg)
-#define LOGI(target, msg) ::logx::emit(::logx::Level::Info, target, msg)
-#define LOGW(target, msg) ::logx::emit(::logx::Level::Warn, target, msg)
-#define LOGE(target, msg) ::logx::emit(::logx::Level::Error, target, msg)
+// The message expression is only evaluated when the level is enabled — hot
+// loops log per pod, and the string concatenation would otherwise dominate
+// filtered-out levels.
+#define LOGX_AT(lvl, target, msg)                                   \
+  do {                                                              \
+    if (::logx::enabled(lvl)) ::logx::emit(lvl, target, msg);       \
+  } while (0)
+#define LOGT(target, msg) LOGX_AT(::logx::Level::Trace, target, msg)
+#define LOGD(target, msg) LOGX_AT(::logx::Level::Debug, target, msg)
+#define LOGI(target, msg) LOGX_AT(::logx::Level::Info, target, msg)
+#define LOGW(target, msg) LOGX_AT(::logx::Level::Warn, target, msg)
+#define LOGE(target, msg) LOGX_AT(::logx::Level::Error, target, msg)
 
 // ---- counters ---------------------------------------------------------------
 // Named monotonic counters and gauges, mirroring the reference's
