@@ -237,6 +237,19 @@ PYBIND11_MODULE(_pruner_core, m) {
 
   m.def("get_prometheus_token", [] { return get_prometheus_token(); });
 
+  m.def("resolve_kube_config", [] {
+    KubeConfig cfg = KubeConfig::resolve();
+    py::dict d;
+    d["url"] = cfg.url;
+    d["token"] = cfg.token ? py::cast(*cfg.token) : py::object(py::none());
+    d["token_file"] =
+        cfg.token_file ? py::cast(*cfg.token_file) : py::object(py::none());
+    d["ca_file"] = cfg.ca_file ? py::cast(*cfg.ca_file) : py::object(py::none());
+    d["skip_tls"] = cfg.skip_tls;
+    d["default_namespace"] = cfg.default_namespace;
+    return d;
+  });
+
   // ---- native synthetic backend (benchmark harness) ----
   py::class_<SyntheticBackend>(m, "SyntheticBackend")
       .def(py::init([](int n_pods, int pods_per_parent, int gpus_per_pod, int latency_us,

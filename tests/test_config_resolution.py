@@ -1,0 +1,95 @@
+"""Kubernetes config + Prometheus token-chain resolution tests.
+
+Pins the in-cluster config path (KUBERNETES_SERVICE_HOST + service-account
+dir: token file, ca.crt, namespace — what kube-rs Config::infer does for the
+reference) and the Prometheus token chain ($PROMETHEUS_TOKEN → SA token file
+→ K8s token env → `oc whoami -t`, reference lib.rs:205-230).
+"""
+
+import pytest
+
+
+@pytest.fixture
+def sa_dir(tmp_path):
+    d = tmp_path / "serviceaccount"
+    d.mkdir()
+    (d / "token").write_text("sa-token-123\n")
+    (d / "ca.crt").write_text("-----BEGIN CERTIFICATE-----\nfake\n-----END CERTIFICATE-----\n")
+    (d / "namespace").write_text("gpu-pruner-system")
+    return d
+
+
+@pytest.fixture
+def clean_env(monkeypatch):
+    for var in ("GPU_PRUNER_K8S_URL", "GPU_PRUNER_K8S_TOKEN",
+                "GPU_PRUNER_K8S_TOKEN_FILE", "GPU_PRUNER_K8S_CA",
+                "GPU_PRUNER_K8S_SKIP_TLS", "KUBERNETES_SERVICE_HOST",
+                "KUBERNETES_SERVICE_PORT", "GPU_PRUNER_SA_DIR",
+                "PROMETHEUS_TOKEN"):
+        monkeypatch.delenv(var, raising=False)
+    return monkeypatch
+
+
+def test_in_cluster_config(core, clean_env, sa_dir):
+    clean_env.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    clean_env.setenv("KUBERNETES_SERVICE_PORT", "6443")
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(sa_dir))
+    cfg = core.resolve_kube_config()
+    assert cfg["url"] == "https://10.0.0.1:6443"
+    assert cfg["token_file"] == str(sa_dir / "token")
+    assert cfg["ca_file"] == str(sa_dir / "ca.crt")
+    assert cfg["default_namespace"] == "gpu-pruner-system"
+    assert cfg["skip_tls"] is False
+
+
+def test_in_cluster_ipv6_host(core, clean_env, sa_dir):
+    clean_env.setenv("KUBERNETES_SERVICE_HOST", "fd00::1")
+    clean_env.setenv("KUBERNETES_SERVICE_PORT", "443")
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(sa_dir))
+    cfg = core.resolve_kube_config()
+    assert cfg["url"] == "https://[fd00::1]:443"
+
+
+def test_env_override_takes_precedence(core, clean_env, sa_dir):
+    clean_env.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(sa_dir))
+    clean_env.setenv("GPU_PRUNER_K8S_URL", "http://127.0.0.1:8080")
+    clean_env.setenv("GPU_PRUNER_K8S_TOKEN", "override-token")
+    cfg = core.resolve_kube_config()
+    assert cfg["url"] == "http://127.0.0.1:8080"
+    assert cfg["token"] == "override-token"
+
+
+def test_no_config_raises(core, clean_env):
+    with pytest.raises(RuntimeError, match="no Kubernetes config"):
+        core.resolve_kube_config()
+
+
+# ---- Prometheus token chain -------------------------------------------------
+
+
+def test_prom_token_env_first(core, clean_env, sa_dir):
+    clean_env.setenv("PROMETHEUS_TOKEN", "env-tok")
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(sa_dir))
+    assert core.get_prometheus_token() == "env-tok"
+
+
+def test_prom_token_sa_file_second(core, clean_env, sa_dir):
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(sa_dir))
+    assert core.get_prometheus_token() == "sa-token-123"
+
+
+def test_prom_token_k8s_env_third(core, clean_env, tmp_path):
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(tmp_path / "nonexistent"))
+    clean_env.setenv("GPU_PRUNER_K8S_TOKEN", "k8s-tok")
+    assert core.get_prometheus_token() == "k8s-tok"
+
+
+def test_prom_token_oc_whoami_last(core, clean_env, tmp_path, monkeypatch):
+    """Last resort shells out to `oc whoami -t`."""
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(tmp_path / "nonexistent"))
+    oc = tmp_path / "oc"
+    oc.write_text("#!/bin/sh\necho oc-token-456\n")
+    oc.chmod(0o755)
+    monkeypatch.setenv("PATH", f"{tmp_path}:/usr/bin:/bin")
+    assert core.get_prometheus_token() == "oc-token-456"
