@@ -62,6 +62,15 @@ std::string render_metrics(const std::vector<DeviceSample>& samples,
        [](const DeviceSample& d) { return d.temp_edge_c; }},
       {"DCGM_FI_DEV_SM_CLOCK", "Graphics clock (MHz)", "gauge",
        [](const DeviceSample& d) { return d.gfx_clock_mhz; }},
+      // AMD-native extras: xGMI topology + accumulated link traffic
+      {"mi355_xgmi_link_width", "xGMI link width (lanes)", "gauge",
+       [](const DeviceSample& d) { return d.xgmi_link_width; }},
+      {"mi355_xgmi_link_speed", "xGMI link speed", "gauge",
+       [](const DeviceSample& d) { return d.xgmi_link_speed; }},
+      {"mi355_xgmi_read_kb_total", "Accumulated xGMI reads across links (KiB)",
+       "counter", [](const DeviceSample& d) { return d.xgmi_read_kb; }},
+      {"mi355_xgmi_write_kb_total", "Accumulated xGMI writes across links (KiB)",
+       "counter", [](const DeviceSample& d) { return d.xgmi_write_kb; }},
   };
 
   std::string out;

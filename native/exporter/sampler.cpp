@@ -180,6 +180,15 @@ void Sampler::poll_device(uint32_t i) {
     if (gm.current_socket_power) d.power_w = gm.current_socket_power;
     // energy_accumulator counts 15.259 µJ per unit
     d.energy_j = static_cast<double>(gm.energy_accumulator) * 15.259e-6;
+    d.xgmi_link_width = gm.xgmi_link_width;
+    d.xgmi_link_speed = gm.xgmi_link_speed;
+    double rd = 0, wr = 0;
+    for (int l = 0; l < RSMI_MAX_NUM_XGMI_LINKS; l++) {
+      rd += static_cast<double>(gm.xgmi_read_data_acc[l]);
+      wr += static_cast<double>(gm.xgmi_write_data_acc[l]);
+    }
+    d.xgmi_read_kb = rd;
+    d.xgmi_write_kb = wr;
     // prefer the firmware activity percentage when the busy-percent sysfs
     // read is unsupported
     if (d.busy_percent == 0.0 && gm.average_gfx_activity > 0 &&

@@ -45,7 +45,15 @@ struct DeviceSample {
   double energy_j = 0.0;           // accumulated energy (J) when supported
   bool metrics_table_ok = false;   // gpu_metrics table was readable
   uint64_t gfx_activity_acc = 0;   // raw accumulated activity counter
-  uint64_t firmware_timestamp = 0; // 10 ns units
+  uint64_t firmware_timestamp = 0; // ns on gfx950 (header says 10 ns)
+
+  // xGMI topology/traffic (topology awareness only — this
+  // daemon moves no tensors): per-device link width/speed and the firmware's
+  // accumulated per-link transfer counters, summed (KiB units per rocm_smi).
+  double xgmi_link_width = 0.0;
+  double xgmi_link_speed = 0.0;
+  double xgmi_read_kb = 0.0;
+  double xgmi_write_kb = 0.0;
 };
 
 class SamplerError : public std::runtime_error {

@@ -39,6 +39,10 @@ py::dict sample_to_dict(const DeviceSample& d) {
   s["metrics_table_ok"] = d.metrics_table_ok;
   s["gfx_activity_acc"] = d.gfx_activity_acc;
   s["firmware_timestamp"] = d.firmware_timestamp;
+  s["xgmi_link_width"] = d.xgmi_link_width;
+  s["xgmi_link_speed"] = d.xgmi_link_speed;
+  s["xgmi_read_kb"] = d.xgmi_read_kb;
+  s["xgmi_write_kb"] = d.xgmi_write_kb;
   return s;
 }
 
