@@ -8,6 +8,7 @@
 
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
+#include "../common/threadpool.hpp"
 
 namespace pruner {
 
@@ -133,36 +134,23 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
 size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
                  uint8_t enabled_mask, int concurrency) {
   std::atomic<size_t> scaled{0};
-  std::atomic<size_t> next{0};
-  auto worker = [&]() {
-    while (true) {
-      size_t i = next.fetch_add(1, std::memory_order_relaxed);
-      if (i >= roots.size()) break;
-      const ScaleKind& sk = roots[i];
-      if (!(enabled_mask & kind_flag(sk.kind))) {
-        LOGI(TARGET, "Skipping resource type " + sk.kind_str() + " because it is not enabled");
-        continue;
-      }
-      try {
-        scale(kube, sk);
-        scaled.fetch_add(1, std::memory_order_relaxed);
-        logx::counter_add("monotonic_counter.scale_successes", 1);
-      } catch (const std::exception& e) {
-        logx::counter_add("monotonic_counter.scale_failures", 1);
-        LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
-      }
-    }
-  };
-  size_t n_workers = std::min<size_t>(static_cast<size_t>(std::max(concurrency, 1)),
-                                      roots.size());
-  if (n_workers <= 1) {
-    worker();
-  } else {
-    std::vector<std::thread> threads;
-    threads.reserve(n_workers);
-    for (size_t t = 0; t < n_workers; t++) threads.emplace_back(worker);
-    for (auto& t : threads) t.join();
-  }
+  qx::ThreadPool::global().parallel_for(
+      roots.size(), concurrency, [&](size_t i) {
+        const ScaleKind& sk = roots[i];
+        if (!(enabled_mask & kind_flag(sk.kind))) {
+          LOGI(TARGET,
+               "Skipping resource type " + sk.kind_str() + " because it is not enabled");
+          return;
+        }
+        try {
+          scale(kube, sk);
+          scaled.fetch_add(1, std::memory_order_relaxed);
+          logx::counter_add("monotonic_counter.scale_successes", 1);
+        } catch (const std::exception& e) {
+          logx::counter_add("monotonic_counter.scale_failures", 1);
+          LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
+        }
+      });
   return scaled.load();
 }
 
@@ -197,54 +185,40 @@ std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& r
   // Evaluate pods concurrently: each needs 1-3 apiserver round-trips (pod GET
   // + owner walk). Worker count is the --max-concurrency knob.
   std::vector<std::optional<ScaleKind>> results(unique_pods.size());
-  std::atomic<size_t> next{0};
-  auto worker = [&]() {
-    while (true) {
-      size_t i = next.fetch_add(1, std::memory_order_relaxed);
-      if (i >= unique_pods.size()) break;
+  qx::ThreadPool::global().parallel_for(
+      unique_pods.size(), cfg.max_concurrency, [&](size_t i) {
       const PodMetricData& pmd = unique_pods[i];
       try {
         auto pod = kube.get_pod(pmd.ns, pmd.name);
         if (!pod) {
           LOGI(TARGET, "Skipping " + pmd.ns + ":" + pmd.name + ", pod no longer exists");
-          continue;
+          return;
         }
         std::string phase = pod->at({"status", "phase"}).as_string_or("Unknown");
         if (phase == "Pending") {
           LOGI(TARGET, "Skipping pod " + pmd.ns + ":" + pmd.name + ", it's still pending");
-          continue;
+          return;
         }
         const jsn::Value& created = pod->at({"metadata", "creationTimestamp"});
         if (!created.is_string()) {
           LOGW(TARGET,
                "Pod " + pmd.ns + ":" + pmd.name + " has no creation timestamp, skipping");
-          continue;
+          return;
         }
         double created_s = 0;
         if (!strutil::parse_rfc3339(created.as_string(), &created_s)) {
           LOGW(TARGET, "Pod " + pmd.ns + ":" + pmd.name +
                            " has unparseable creation timestamp, skipping");
-          continue;
+          return;
         }
-        if (created_s >= lookback_start) continue;  // too young for the window
+        if (created_s >= lookback_start) return;  // too young for the window
         LOGI(TARGET, "Pod " + pmd.ns + ":" + pmd.name + " is idle and eligible for scaledown");
         results[i] = find_root_object(kube, *pod);
       } catch (const std::exception& e) {
         LOGE(TARGET,
              "Skipping " + pmd.ns + ":" + pmd.name + ", retrieval error: " + e.what());
       }
-    }
-  };
-  size_t n_workers =
-      std::min<size_t>(static_cast<size_t>(cfg.max_concurrency), unique_pods.size());
-  if (n_workers <= 1) {
-    worker();
-  } else {
-    std::vector<std::thread> threads;
-    threads.reserve(n_workers);
-    for (size_t t = 0; t < n_workers; t++) threads.emplace_back(worker);
-    for (auto& t : threads) t.join();
-  }
+      });
 
   // Pods sharing a parent collapse to one scale action (uid-hash dedup,
   // reference main.rs:534).

@@ -13,6 +13,7 @@
 
 #include "../common/json.hpp"
 #include "../common/queue.hpp"
+#include "../common/threadpool.hpp"
 #include "../common/strutil.hpp"
 #include "../pruner/config.hpp"
 #include "../pruner/promql.hpp"
@@ -105,6 +106,24 @@ static void test_queue_mpmc() {
   CHECK(sum.load() == expect);
 }
 
+static void test_threadpool() {
+  // TSan target: bounded parallel_for correctness incl. caller participation
+  std::atomic<long> sum{0};
+  qx::ThreadPool pool(8);
+  pool.parallel_for(1000, 16, [&](size_t i) { sum.fetch_add(static_cast<long>(i)); });
+  CHECK(sum.load() == 999 * 1000 / 2);
+  sum = 0;
+  pool.parallel_for(5, 1, [&](size_t i) { sum.fetch_add(static_cast<long>(i) + 1); });
+  CHECK(sum.load() == 15);
+  pool.parallel_for(0, 8, [&](size_t) { CHECK(false); });
+  // repeated use reuses the same workers
+  for (int round = 0; round < 50; round++) {
+    std::atomic<int> n{0};
+    pool.parallel_for(64, 32, [&](size_t) { n.fetch_add(1); });
+    CHECK(n.load() == 64);
+  }
+}
+
 static void test_cli() {
   auto r = parse_cli({"--prometheus-url", "http://p", "-t", "15", "--run-mode",
                       "scale-down"});
@@ -132,6 +151,7 @@ int main() {
   test_promql();
   test_resources();
   test_queue_mpmc();
+  test_threadpool();
   test_cli();
   test_strutil();
   if (failures == 0) std::puts("native unit tests: all passed");
