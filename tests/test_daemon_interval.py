@@ -1,0 +1,61 @@
+"""Daemon-loop cadence + scale-consumer tests."""
+
+import os
+import subprocess
+import time
+
+import pytest
+
+
+def test_daemon_mode_ticks_on_interval(pruner_bin, fake_api, fake_prom):
+    """--check-interval=1: at least 2 and at most ~5 queries in ~3.5s."""
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    p = subprocess.Popen(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
+         "--check-interval", "1"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        time.sleep(3.5)
+        assert p.poll() is None, "daemon should keep running"
+        n = len(fake_prom.queries)
+        assert 2 <= n <= 6, f"expected ~3 ticks at 1s interval, saw {n}"
+    finally:
+        p.kill()
+        p.wait()
+
+
+def test_one_shot_mode_runs_once_and_exits(pruner_bin, fake_api, fake_prom):
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run([pruner_bin, "--prometheus-url", fake_prom.url],
+                       capture_output=True, text=True, timeout=30, env=env)
+    assert r.returncode == 0
+    assert len(fake_prom.queries) == 1
+
+
+@pytest.mark.parametrize("n_pods", [200])
+def test_large_cluster_scale_down_complete(pruner_bin, fake_api, fake_prom, n_pods):
+    """Stress shape (BASELINE config 5, scaled down for CI): every parent of
+    an idle pod is scaled exactly once and announced exactly once."""
+    from gpu_pruner_amd.fixtures import build_synthetic_cluster
+
+    info = build_synthetic_cluster(fake_api, fake_prom, n_pods=n_pods,
+                                   pods_per_parent=4)
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--run-mode", "scale-down",
+         "--max-concurrency", "64"],
+        capture_output=True, text=True, timeout=120, env=env)
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert len(fake_api.events) == info["expected_shutdown_events"]
+    scaled = sum(
+        1 for (kind, _, _), obj in fake_api.objects.items()
+        if (kind in ("Deployment", "StatefulSet") and obj["spec"].get("replicas") == 0)
+        or (kind == "Notebook" and "kubeflow-resource-stopped" in obj["metadata"].get("annotations", {}))
+        or (kind == "InferenceService" and obj["spec"]["predictor"].get("minReplicas") == 0))
+    assert scaled == info["expected_shutdown_events"]
