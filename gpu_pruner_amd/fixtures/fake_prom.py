@@ -22,6 +22,9 @@ class FakePrometheus:
         self.queries: list[str] = []
         self.fail_next = 0  # serve this many 500s before succeeding
         self.bearer_tokens: list[str | None] = []
+        # when set, served verbatim as the API `data` object (e.g. a matrix
+        # result for querytest tests)
+        self.data_override: dict | None = None
 
         fixture = self
 
@@ -43,14 +46,12 @@ class FakePrometheus:
                         fixture.fail_next -= 1
                         self._send(500, {"status": "error", "error": "induced failure"})
                         return
-                    result = [dict(s) for s in fixture.series]
-                self._send(
-                    200,
-                    {
-                        "status": "success",
-                        "data": {"resultType": "vector", "result": result},
-                    },
-                )
+                    if fixture.data_override is not None:
+                        data = dict(fixture.data_override)
+                    else:
+                        data = {"resultType": "vector",
+                                "result": [dict(s) for s in fixture.series]}
+                self._send(200, {"status": "success", "data": data})
 
             def _send(self, status: int, obj: dict):
                 body = json.dumps(obj).encode()

@@ -9,6 +9,7 @@
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
 #include "../common/threadpool.hpp"
+#include "otlp.hpp"
 
 namespace pruner {
 
@@ -17,6 +18,7 @@ constexpr const char* TARGET = "pruner::engine";
 }
 
 std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& pod) {
+  otlp::SpanGuard span("find_root_object");
   const jsn::Value& meta = pod.get("metadata");
   std::string pod_name = meta.get("name").as_string();
   std::string ns = meta.get("namespace").as_string_or("");
@@ -88,6 +90,7 @@ std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& po
 }
 
 void scale(KubeClient& kube, const ScaleKind& sk) {
+  otlp::SpanGuard span("scale");
   auto ns = sk.ns();
   if (ns) {
     // Announce first; a failed Event post never blocks the scale itself
@@ -107,6 +110,7 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
     case Kind::ReplicaSet:
     case Kind::StatefulSet: {
       // Built-in workloads: spec.replicas=0 through the /scale subresource.
+      otlp::SpanGuard sub("scale_to_zero");
       jsn::Value patch = jsn::Value::object();
       patch["spec"]["replicas"] = 0;
       kube.patch_scale(sk.kind, namespace_, sk.name(), patch);
@@ -114,6 +118,7 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
     }
     case Kind::Notebook: {
       // Kubeflow convention: the stop annotation, set to "now".
+      otlp::SpanGuard sub("scale_notebook_to_zero");
       jsn::Value patch = jsn::Value::object();
       patch["metadata"]["annotations"]["kubeflow-resource-stopped"] = strutil::rfc3339_now();
       kube.merge_patch(object_path(Kind::Notebook, namespace_, sk.name()), patch);
@@ -123,6 +128,7 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
       // KServe scales the predictor down itself once minReplicas is 0 and
       // rescales on traffic; durable capacity needs a manual minReplicas
       // reset (same semantics as the reference, lib.rs:553-576).
+      otlp::SpanGuard sub("scale_inference_service_to_zero");
       jsn::Value patch = jsn::Value::object();
       patch["spec"]["predictor"]["minReplicas"] = 0;
       kube.merge_patch(object_path(Kind::InferenceService, namespace_, sk.name()), patch);

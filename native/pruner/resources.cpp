@@ -4,6 +4,7 @@
 #include <functional>
 
 #include "../common/strutil.hpp"
+#include "otlp.hpp"
 
 namespace pruner {
 
@@ -93,6 +94,7 @@ size_t ScaleKind::hash() const {
 }
 
 jsn::Value generate_scale_event(const ScaleKind& sk) {
+  otlp::SpanGuard span("generate_scale_event");
   std::string now = strutil::rfc3339_now();
   std::string now_micro = strutil::rfc3339_micro_now();
 

@@ -44,8 +44,13 @@ def test_otlp_spans_and_counters_exported(pruner_bin, idle_cluster, fake_prom):
                        env_extra={"OTEL_EXPORTER_OTLP_ENDPOINT": collector.url,
                                   "OTEL_METRIC_EXPORT_INTERVAL": "60000"})
         assert r.returncode == 0, r.stderr
-        # shutdown() final-flushes, so one-shot runs still export
-        assert "run_query_and_scale" in collector.span_names()
+        # shutdown() final-flushes, so one-shot runs still export. The span
+        # surface matches the reference's 7 #[tracing::instrument] sites
+        # (SURVEY.md §5.1).
+        names = collector.span_names()
+        for span in ("run_query_and_scale", "find_root_object", "scale",
+                     "generate_scale_event", "scale_to_zero"):
+            assert span in names, f"missing span {span} in {set(names)}"
         points = collector.metric_points()
         assert points.get("query_successes") == 1
         assert points.get("scale_successes") == 1

@@ -111,3 +111,35 @@ def test_querytest_requires_args(querytest_bin):
     r = subprocess.run([querytest_bin], capture_output=True, text=True, timeout=10)
     assert r.returncode == 2
     assert "usage" in r.stderr
+
+
+def test_querytest_matrix_result(querytest_bin, fake_prom, tmp_path):
+    """Range-vector (matrix) results print one row per sample."""
+    fake_prom.data_override = {
+        "resultType": "matrix",
+        "result": [{
+            "metric": {"pod": "pod-m", "namespace": "ml"},
+            "values": [[1700000000, "0.1"], [1700000060, "0.2"]],
+        }],
+    }
+    env = dict(os.environ)
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run(
+        [querytest_bin, "rate(x[5m])", fake_prom.url],
+        capture_output=True, text=True, timeout=30, env=env, cwd=str(tmp_path))
+    assert r.returncode == 0, r.stderr
+    assert r.stdout.count("pod-m") == 2
+    assert "0.1" in r.stdout and "0.2" in r.stdout
+    rows = list(csv.reader((tmp_path / "output.csv").open()))
+    assert len(rows) == 2
+
+
+def test_querytest_scalar_unsupported(querytest_bin, fake_prom):
+    """Scalar results are rejected (parity with reference querytest.rs:54-56)."""
+    fake_prom.data_override = {"resultType": "scalar", "result": [1700000000, "1"]}
+    env = dict(os.environ)
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run([querytest_bin, "1", fake_prom.url],
+                       capture_output=True, text=True, timeout=30, env=env)
+    assert r.returncode == 1
+    assert "Scalar data not supported" in r.stderr
