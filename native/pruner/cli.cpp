@@ -42,6 +42,8 @@ OPTIONS:
       --queue-capacity <N>        scale-event queue bound [default: 100]
       --max-failures <N>          abort after more than N consecutive query
                                   failures [default: 5]
+      --metrics-port <PORT>       serve daemon self-metrics + /healthz on
+                                  this port (0 = disabled) [default: 0]
   -h, --help                      print this help
 )";
 
@@ -80,6 +82,7 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
       {"--max-concurrency", "max-concurrency"},
       {"--queue-capacity", "queue-capacity"},
       {"--max-failures", "max-failures"},
+      {"--metrics-port", "metrics-port"},
       {"-h", "help"},              {"--help", "help"},
   };
   // flags that never take a value
@@ -167,6 +170,7 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
       } else if (name == "max-concurrency") c.max_concurrency = std::stoi(val);
       else if (name == "queue-capacity") c.queue_capacity = std::stoi(val);
       else if (name == "max-failures") c.max_consecutive_failures = std::stoi(val);
+      else if (name == "metrics-port") c.metrics_port = std::stoi(val);
     } catch (const std::exception&) {
       return fail("invalid value for --" + name + ": " + val);
     }

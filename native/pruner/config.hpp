@@ -40,6 +40,7 @@ struct Config {
   int max_concurrency = 32;        // --max-concurrency
   int queue_capacity = 100;        // --queue-capacity
   int max_consecutive_failures = 5;  // --max-failures (abort after more than N)
+  int metrics_port = 0;            // --metrics-port (0 = disabled)
 
   QueryArgs query_args() const {
     QueryArgs qa;

@@ -1,12 +1,17 @@
 #include "daemon.hpp"
 
+#include <csignal>
+
 #include <atomic>
 #include <chrono>
+#include <condition_variable>
 #include <memory>
 #include <mutex>
+#include <sstream>
 #include <thread>
 #include <vector>
 
+#include "../common/http_server.hpp"
 #include "../common/log.hpp"
 #include "../common/queue.hpp"
 #include "engine.hpp"
@@ -17,7 +22,34 @@ namespace pruner {
 
 namespace {
 constexpr const char* TARGET = "pruner::daemon";
+
+// Graceful shutdown: SIGTERM/SIGINT set the flag; the interval wait is a
+// condition-variable sleep so K8s pod termination interrupts a 180 s tick
+// wait immediately, finishes in-flight scale actions, and exits 0.
+std::atomic<bool> g_shutdown{false};
+std::mutex g_shutdown_mu;
+std::condition_variable g_shutdown_cv;
+
+void on_shutdown_signal(int) {
+  g_shutdown.store(true);
+  g_shutdown_cv.notify_all();
 }
+
+// Daemon self-metrics in Prometheus text format (--metrics-port): the six
+// counters of SURVEY.md §5.5 plus a liveness endpoint.
+std::string render_self_metrics() {
+  std::ostringstream out;
+  for (const auto& [name, value] : logx::counters_snapshot()) {
+    bool monotonic = name.rfind("monotonic_counter.", 0) == 0;
+    std::string short_name = "gpu_pruner_" + name.substr(name.find('.') + 1);
+    if (monotonic) short_name += "_total";
+    out << "# TYPE " << short_name << (monotonic ? " counter" : " gauge") << "\n";
+    out << short_name << " " << value << "\n";
+  }
+  return out.str();
+}
+
+}  // namespace
 
 int run_daemon(const Config& cfg) {
   const uint8_t enabled = get_enabled_resources(cfg.enabled_resources);
@@ -71,11 +103,39 @@ int run_daemon(const Config& cfg) {
   consumers.reserve(static_cast<size_t>(n_consumers));
   for (int i = 0; i < n_consumers; i++) consumers.emplace_back(consume);
 
+  g_shutdown.store(false);
+  std::signal(SIGTERM, on_shutdown_signal);
+  std::signal(SIGINT, on_shutdown_signal);
+
+  // Optional self-metrics endpoint (not part of the reference's surface).
+  std::unique_ptr<http::Server> metrics_server;
+  if (cfg.metrics_port > 0) {
+    metrics_server = std::make_unique<http::Server>(
+        "0.0.0.0", static_cast<uint16_t>(cfg.metrics_port),
+        [](const http::ServerRequest& req) {
+          http::ServerResponse resp;
+          if (req.path == "/metrics") {
+            resp.body = render_self_metrics();
+            resp.content_type = "text/plain; version=0.0.4; charset=utf-8";
+          } else if (req.path == "/healthz") {
+            resp.body = "ok\n";
+          } else {
+            resp.status = 404;
+            resp.body = "not found\n";
+          }
+          return resp;
+        });
+    metrics_server->start();
+    LOGI(TARGET, "Self-metrics on :" + std::to_string(metrics_server->port()));
+  }
+
   int consecutive_failures = 0;
   auto next_tick = std::chrono::steady_clock::now();
-  while (true) {
+  while (!g_shutdown.load()) {
     if (cfg.daemon_mode) {
-      std::this_thread::sleep_until(next_tick);
+      std::unique_lock<std::mutex> lock(g_shutdown_mu);
+      g_shutdown_cv.wait_until(lock, next_tick, [] { return g_shutdown.load(); });
+      if (g_shutdown.load()) break;
       next_tick += std::chrono::seconds(cfg.check_interval_s);
     }
     try {
@@ -106,8 +166,10 @@ int run_daemon(const Config& cfg) {
     if (!cfg.daemon_mode) break;
   }
 
+  if (g_shutdown.load()) LOGI(TARGET, "Shutdown signal received, draining");
   queue.close();  // producer done: consumers drain and exit
   for (auto& c : consumers) c.join();
+  if (metrics_server) metrics_server->stop();
   return exit_code.load();
 }
 

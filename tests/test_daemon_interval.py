@@ -59,3 +59,67 @@ def test_large_cluster_scale_down_complete(pruner_bin, fake_api, fake_prom, n_po
         or (kind == "Notebook" and "kubeflow-resource-stopped" in obj["metadata"].get("annotations", {}))
         or (kind == "InferenceService" and obj["spec"]["predictor"].get("minReplicas") == 0))
     assert scaled == info["expected_shutdown_events"]
+
+
+def test_sigterm_graceful_shutdown(pruner_bin, fake_api, fake_prom):
+    """K8s pod termination: SIGTERM interrupts the tick wait, drains, exits 0."""
+    import signal
+
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    p = subprocess.Popen(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
+         "--check-interval", "180"],  # long interval: the wait must be interruptible
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        time.sleep(1.0)
+        assert p.poll() is None
+        t0 = time.monotonic()
+        p.send_signal(signal.SIGTERM)
+        rc = p.wait(timeout=10)
+        assert time.monotonic() - t0 < 5, "shutdown should not wait out the interval"
+        assert rc == 0
+        assert b"draining" in p.stderr.read()
+    finally:
+        if p.poll() is None:
+            p.kill()
+            p.wait()
+
+
+def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
+    """--metrics-port serves the six counters + /healthz (MI355X-native add)."""
+    import urllib.request
+
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    fake_api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p0", "ml")
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    p = subprocess.Popen(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
+         "--check-interval", "1", "--run-mode", "scale-down",
+         "--metrics-port", "19490"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        text = None
+        for _ in range(40):
+            time.sleep(0.2)
+            try:
+                text = urllib.request.urlopen(
+                    "http://127.0.0.1:19490/metrics", timeout=2).read().decode()
+                if "gpu_pruner_query_successes_total" in text:
+                    break
+            except OSError:
+                continue
+        assert text and "gpu_pruner_query_successes_total" in text
+        assert "gpu_pruner_scale_successes_total" in text
+        health = urllib.request.urlopen(
+            "http://127.0.0.1:19490/healthz", timeout=2).read()
+        assert health == b"ok\n"
+    finally:
+        p.terminate()
+        p.wait(timeout=10)
