@@ -62,7 +62,7 @@ void SyntheticBackend::build_cluster() {
       d["kind"] = "Deployment";
       d["metadata"] = meta(dep, ns, "Deployment");
       d["spec"]["replicas"] = 1;
-      objects_["Deployment"][ns][dep] = d;
+      objects_["Deployment"][ns][dep] = {d, d.dump()};
       jsn::Value r = jsn::Value::object();
       r["apiVersion"] = "apps/v1";
       r["kind"] = "ReplicaSet";
@@ -74,7 +74,7 @@ void SyntheticBackend::build_cluster() {
       owner["uid"] = uid_for("Deployment", ns, dep);
       r["metadata"]["ownerReferences"] = jsn::Value(jsn::Array{owner});
       r["spec"]["replicas"] = 1;
-      objects_["ReplicaSet"][ns][rs] = r;
+      objects_["ReplicaSet"][ns][rs] = {r, r.dump()};
       parents.push_back({"ReplicaSet", rs, ns});
     } else if (flavor == 1) {
       std::string nb = "nb-" + std::to_string(p);
@@ -84,7 +84,7 @@ void SyntheticBackend::build_cluster() {
       n["kind"] = "Notebook";
       n["metadata"] = meta(nb, ns, "Notebook");
       n["spec"]["template"] = nullptr;
-      objects_["Notebook"][ns][nb] = n;
+      objects_["Notebook"][ns][nb] = {n, n.dump()};
       jsn::Value s = jsn::Value::object();
       s["apiVersion"] = "apps/v1";
       s["kind"] = "StatefulSet";
@@ -96,7 +96,7 @@ void SyntheticBackend::build_cluster() {
       owner["uid"] = uid_for("Notebook", ns, nb);
       s["metadata"]["ownerReferences"] = jsn::Value(jsn::Array{owner});
       s["spec"]["replicas"] = 1;
-      objects_["StatefulSet"][ns][ss] = s;
+      objects_["StatefulSet"][ns][ss] = {s, s.dump()};
       parents.push_back({"StatefulSet", ss, ns});
     } else {
       std::string isvc = "isvc-" + std::to_string(p);
@@ -105,7 +105,7 @@ void SyntheticBackend::build_cluster() {
       v["kind"] = "InferenceService";
       v["metadata"] = meta(isvc, ns, "InferenceService");
       v["spec"]["predictor"]["minReplicas"] = 1;
-      objects_["InferenceService"][ns][isvc] = v;
+      objects_["InferenceService"][ns][isvc] = {v, v.dump()};
       parents.push_back({"", isvc, ns});
     }
   }
@@ -134,7 +134,7 @@ void SyntheticBackend::build_cluster() {
       p["metadata"]["ownerReferences"] = jsn::Value(jsn::Array{owner});
     }
     p["status"]["phase"] = "Running";
-    objects_["Pod"][par.ns][pod] = p;
+    objects_["Pod"][par.ns][pod] = {p, p.dump()};
 
     for (int g = 0; g < opts_.gpus_per_pod; g++) {
       if (!first) series += ",";
@@ -248,7 +248,7 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
 
   std::lock_guard<std::mutex> lock(mu_);
   auto kit = objects_.find(kind);
-  jsn::Value* obj = nullptr;
+  StoredObject* obj = nullptr;
   if (kit != objects_.end()) {
     auto nit = kit->second.find(ns);
     if (nit != kit->second.end()) {
@@ -262,20 +262,23 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
     return resp;
   }
   if (req.method == "GET") {
-    resp.body = obj->dump();
+    if (obj->cached_dump.empty()) obj->cached_dump = obj->obj.dump();
+    resp.body = obj->cached_dump;
     return resp;
   }
   if (req.method == "PATCH") {
     jsn::Value patch = jsn::parse(req.body);
     if (is_scale) {
       scale_patches_.fetch_add(1, std::memory_order_relaxed);
-      (*obj)["spec"]["replicas"] = patch.at({"spec", "replicas"});
+      obj->obj["spec"]["replicas"] = patch.at({"spec", "replicas"});
+      obj->cached_dump.clear();
       resp.body = "{\"kind\":\"Scale\",\"spec\":" + patch.get("spec").dump() + "}";
       return resp;
     }
     scale_patches_.fetch_add(1, std::memory_order_relaxed);
-    obj->merge_patch(patch);
-    resp.body = obj->dump();
+    obj->obj.merge_patch(patch);
+    obj->cached_dump.clear();
+    resp.body = obj->obj.dump();
     return resp;
   }
   resp.status = 405;

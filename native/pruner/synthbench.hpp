@@ -56,9 +56,13 @@ private:
 
   SynthOptions opts_;
   int n_parents_ = 0;
+  struct StoredObject {
+    jsn::Value obj;
+    std::string cached_dump;  // invalidated on PATCH; GETs dominate 6:1
+  };
   std::mutex mu_;
   // kind → ns → name → object
-  std::map<std::string, std::map<std::string, std::map<std::string, jsn::Value>>> objects_;
+  std::map<std::string, std::map<std::string, std::map<std::string, StoredObject>>> objects_;
   std::string series_json_zero_;  // pre-rendered result vector (values patched in)
   std::atomic<double> series_value_{0.0};
   std::atomic<int64_t> events_posted_{0};
