@@ -10,6 +10,10 @@
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
 #include "../pruner/k8s.hpp"
+#include "podresources.hpp"
+#include "sampler.hpp"
+
+#include <sys/stat.h>
 
 namespace exporter {
 
@@ -181,6 +185,82 @@ std::map<uint32_t, PodAttribution> Attributor::resolve(
         break;  // first attributed pid wins for this GPU
       }
     }
+  }
+  return out;
+}
+
+}  // namespace exporter
+
+// ---- PodResources-based allocation attribution -----------------------------
+// (kept at the end of the file: depends on sampler.hpp's DeviceSample)
+
+
+namespace exporter {
+
+namespace {
+
+// AMD device-plugin device IDs vary by plugin version/config; match a
+// reported id against every stable identity the sampler knows for a device.
+bool device_id_matches(const std::string& raw_id, const DeviceSample& d) {
+  std::string id = strutil::lower(strutil::trim(raw_id));
+  if (id.empty()) return false;
+  if (id == std::to_string(d.kfd_gpu_id)) return true;
+  std::string uid = strutil::lower(d.unique_id);
+  if (!uid.empty() && (id == uid || id == "0x" + uid)) return true;
+  if (id == "renderd" + std::to_string(d.drm_render_minor)) return true;
+  if (id == "/dev/dri/renderd" + std::to_string(d.drm_render_minor)) return true;
+  // card index convention: render minor 128+N ↔ cardN
+  if (d.drm_render_minor >= 128 &&
+      id == "card" + std::to_string(d.drm_render_minor - 128))
+    return true;
+  if (id == strutil::lower(d.pci_bdf)) return true;
+  if (id == std::to_string(d.index)) return true;
+  return false;
+}
+
+bool is_gpu_resource(const std::string& resource_name) {
+  return resource_name.find("gpu") != std::string::npos ||
+         resource_name.find("amd.com") != std::string::npos;
+}
+
+}  // namespace
+
+std::map<uint32_t, PodAttribution> Attributor::resolve_full(
+    const std::vector<DeviceSample>& devices) {
+  std::map<uint32_t, PodAttribution> out;
+
+  std::string sock = "/var/lib/kubelet/pod-resources/kubelet.sock";
+  if (const char* env = std::getenv("GPU_EXPORTER_PODRESOURCES_SOCKET"); env && *env)
+    sock = env;
+  struct stat st {};
+  if (::stat(sock.c_str(), &st) == 0) {
+    try {
+      auto entries = list_pod_resources(sock);
+      for (const auto& e : entries) {
+        for (const auto& cd : e.devices) {
+          if (!is_gpu_resource(cd.resource_name)) continue;
+          for (const auto& id : cd.device_ids) {
+            for (const auto& d : devices) {
+              if (out.count(d.index) == 0 && device_id_matches(id, d))
+                out[d.index] = PodAttribution{e.pod, e.ns, e.container};
+            }
+          }
+        }
+      }
+    } catch (const std::exception& ex) {
+      LOGW("exporter::attrib",
+           std::string("PodResources attribution failed, falling back to KFD: ") +
+               ex.what());
+    }
+  }
+
+  // KFD usage-based fallback for devices PodResources did not cover.
+  std::vector<std::pair<uint32_t, uint64_t>> unmatched;
+  for (const auto& d : devices)
+    if (out.count(d.index) == 0) unmatched.emplace_back(d.index, d.kfd_gpu_id);
+  if (!unmatched.empty()) {
+    auto kfd = resolve(unmatched);
+    for (auto& [idx, attr] : kfd) out[idx] = std::move(attr);
   }
   return out;
 }

@@ -38,14 +38,24 @@ std::optional<std::string> pod_uid_from_cgroup(const std::string& cgroup_text);
 // Scan the KFD process registry: kfd_gpu_id → pids with that GPU open.
 std::map<uint64_t, std::vector<int>> kfd_gpu_pids();
 
+struct DeviceSample;  // sampler.hpp
+
 class Attributor {
 public:
   Attributor();
 
   // Resolve attributions for the sampled devices (keyed by device index).
-  // kfd ids come from the sampler's topology mapping.
+  // kfd ids come from the sampler's topology mapping (usage-based KFD path).
   std::map<uint32_t, PodAttribution> resolve(
       const std::vector<std::pair<uint32_t, uint64_t>>& index_to_kfd_id);
+
+  // Full chain: kubelet PodResources allocations first (covers
+  // allocated-but-idle GPUs — the culler's target case), KFD process
+  // registry as fallback for unmatched devices. The PodResources socket is
+  // $GPU_EXPORTER_PODRESOURCES_SOCKET (default
+  // /var/lib/kubelet/pod-resources/kubelet.sock); absent socket → fallback.
+  std::map<uint32_t, PodAttribution> resolve_full(
+      const std::vector<DeviceSample>& devices);
 
   // pod UID → attribution; consults the static map file first, then the
   // apiserver cache (refreshing at most every `refresh_s`).

@@ -108,9 +108,7 @@ int main(int argc, char** argv) {
     http::ServerResponse resp;
     if (req.path == "/metrics") {
       auto samples = sampler.snapshot(/*reset_window=*/true);
-      std::vector<std::pair<uint32_t, uint64_t>> idx_kfd;
-      for (const auto& s : samples) idx_kfd.emplace_back(s.index, s.kfd_gpu_id);
-      auto attribs = attributor.resolve(idx_kfd);
+      auto attribs = attributor.resolve_full(samples);
       resp.body = exporter::render_metrics(samples, attribs, opts);
       resp.content_type = "text/plain; version=0.0.4; charset=utf-8";
     } else if (req.path == "/healthz") {

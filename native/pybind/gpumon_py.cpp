@@ -11,6 +11,7 @@
 #include "../common/json.hpp"
 #include "../common/log.hpp"
 #include "../exporter/attrib.hpp"
+#include "../exporter/podresources.hpp"
 #include "../exporter/registry.hpp"
 #include "../exporter/sampler.hpp"
 
@@ -53,6 +54,7 @@ DeviceSample sample_from_json(const jsn::Value& v) {
   d.unique_id = v.get("unique_id").as_string_or("");
   d.drm_render_minor = static_cast<uint32_t>(v.get("drm_render_minor").as_int(128));
   d.kfd_gpu_id = static_cast<uint64_t>(v.get("kfd_gpu_id").as_int(0));
+  d.pci_bdf = v.get("pci_bdf").as_string_or("");
   d.busy_percent = v.get("busy_percent").as_double(0);
   d.gr_engine_active = v.get("gr_engine_active").as_double(0);
   d.mem_busy_percent = v.get("mem_busy_percent").as_double(0);
@@ -91,6 +93,59 @@ PYBIND11_MODULE(_gpumon, m) {
            },
            py::arg("reset_window") = false);
 
+  m.def("list_pod_resources",
+        [](const std::string& socket_path, int timeout_ms) {
+          std::vector<PodResourcesEntry> entries;
+          {
+            py::gil_scoped_release nogil;
+            entries = list_pod_resources(socket_path, timeout_ms);
+          }
+          py::list out;
+          for (const auto& e : entries) {
+            py::dict d;
+            d["pod"] = e.pod;
+            d["namespace"] = e.ns;
+            d["container"] = e.container;
+            py::list devs;
+            for (const auto& cd : e.devices) {
+              py::dict dd;
+              dd["resource_name"] = cd.resource_name;
+              dd["device_ids"] = cd.device_ids;
+              devs.append(dd);
+            }
+            d["devices"] = devs;
+            out.append(d);
+          }
+          return out;
+        },
+        py::arg("socket_path"), py::arg("timeout_ms") = 5000,
+        "Unary v1.PodResourcesLister/List over the kubelet unix socket "
+        "(hand-rolled h2c + protobuf, no grpc library)");
+
+  py::register_exception<PodResourcesError>(m, "PodResourcesError");
+
+  m.def("decode_list_response", [](py::bytes payload) {
+    std::string data = payload;
+    auto entries = decode_list_response(data);
+    py::list out;
+    for (const auto& e : entries) {
+      py::dict d;
+      d["pod"] = e.pod;
+      d["namespace"] = e.ns;
+      d["container"] = e.container;
+      py::list devs;
+      for (const auto& cd : e.devices) {
+        py::dict dd;
+        dd["resource_name"] = cd.resource_name;
+        dd["device_ids"] = cd.device_ids;
+        devs.append(dd);
+      }
+      d["devices"] = devs;
+      out.append(d);
+    }
+    return out;
+  });
+
   m.def("pod_uid_from_cgroup", [](const std::string& text) -> py::object {
     auto uid = pod_uid_from_cgroup(text);
     return uid ? py::cast(*uid) : py::none();
@@ -123,6 +178,27 @@ PYBIND11_MODULE(_gpumon, m) {
              }
              return out;
            })
+      .def("resolve_full",
+           [](Attributor& a, const std::string& samples_json) {
+             jsn::Value sv = jsn::parse(samples_json);
+             std::vector<DeviceSample> devices;
+             for (const auto& v : sv.arr()) devices.push_back(sample_from_json(v));
+             std::map<uint32_t, PodAttribution> resolved;
+             {
+               py::gil_scoped_release nogil;
+               resolved = a.resolve_full(devices);
+             }
+             py::dict out;
+             for (const auto& [idx, attr] : resolved) {
+               py::dict v;
+               v["pod"] = attr.pod;
+               v["namespace"] = attr.ns;
+               v["container"] = attr.container;
+               out[py::cast(idx)] = v;
+             }
+             return out;
+           },
+           "Allocation-first attribution: PodResources socket, then KFD fallback")
       .def("lookup_uid", [](Attributor& a, const std::string& uid) -> py::object {
         std::optional<PodAttribution> attr;
         {
