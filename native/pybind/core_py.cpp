@@ -17,6 +17,7 @@
 #include "../pruner/prom.hpp"
 #include "../pruner/promql.hpp"
 #include "../pruner/resources.hpp"
+#include "../common/http.hpp"
 #include "../pruner/synthbench.hpp"
 
 namespace py = pybind11;
@@ -274,6 +275,24 @@ PYBIND11_MODULE(_pruner_core, m) {
       .def_property_readonly("scale_patches", &SyntheticBackend::scale_patches)
       .def_property_readonly("requests_served", &SyntheticBackend::requests_served)
       .def_property_readonly("expected_parents", &SyntheticBackend::expected_parents);
+
+  // test helper: raw GET through the native HTTP client (used to pin
+  // chunked / close-delimited / keep-alive decoding against fixture servers)
+  m.def("_http_get",
+        [](const std::string& url) {
+          http::Response resp;
+          {
+            py::gil_scoped_release nogil;
+            resp = http::fetch(url);
+          }
+          py::dict d;
+          d["status"] = resp.status;
+          d["body"] = py::bytes(resp.body);
+          py::dict headers;
+          for (const auto& [k, v] : resp.headers) headers[py::str(k)] = v;
+          d["headers"] = headers;
+          return d;
+        });
 
   m.def("counters_snapshot", [] {
     py::dict d;

@@ -210,3 +210,43 @@ def test_concurrent_evaluation_matches_serial(core, fake_api, concurrency):
     assert out["num_unique_pods"] == n
     assert out["shutdown_events"] == n
     assert sorted(r.name for r in out["roots"]) == sorted(f"d{i}" for i in range(n))
+
+
+def test_owner_precedence_first_scalable_wins(core, fake_api):
+    """Mixed owner refs: unknown kinds are skipped, first scalable resolves
+    (reference lib.rs:458-506 iterates in order)."""
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    pod = fake_api.add_pod("multi-owner", "ml")
+    pod["metadata"]["ownerReferences"] = [
+        {"apiVersion": "batch/v1", "kind": "Job", "name": "some-job", "uid": "j1"},
+        {"apiVersion": "apps/v1", "kind": "ReplicaSet", "name": "d-rs",
+         "uid": rs["metadata"]["uid"]},
+    ]
+    out = evaluate(core, [series("multi-owner", "ml")])
+    assert [r.kind for r in out["roots"]] == ["Deployment"]
+
+
+def test_kserve_label_takes_precedence_over_owners(core, fake_api):
+    """The KServe label shortcut is checked before owner refs (lib.rs:448-456)."""
+    fake_api.add_inferenceservice("llm", "ml")
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    fake_api.add_pod("pred-0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"],
+                     labels={"serving.kserve.io/inferenceservice": "llm"})
+    out = evaluate(core, [series("pred-0", "ml")])
+    assert [r.kind for r in out["roots"]] == ["InferenceService"]
+
+
+def test_repeat_tick_idempotent(core, fake_api):
+    """Re-running the decision over an already-culled cluster is a no-op
+    patch (reference §5.4: stateless between ticks, idempotent)."""
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    fake_api.add_pod("p", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"])
+    result = [series("p", "ml")]
+    for _ in range(3):
+        out = evaluate(core, result)
+        assert out["shutdown_events"] == 1
