@@ -180,8 +180,17 @@ http::ServerResponse SyntheticBackend::handle_prom(const http::ServerRequest& re
     resp.body = "{}";
     return resp;
   }
+  // Emulate the idle query's `== 0` predicate: Prometheus only returns
+  // series whose peak activity over the window is zero, so a non-zero
+  // activity value (busy GPU) yields an empty vector — no candidates.
+  double v = series_value_.load();
+  if (v != 0.0) {
+    resp.content_type = "application/json";
+    resp.body = "{\"status\":\"success\",\"data\":{\"resultType\":\"vector\",\"result\":[]}}";
+    return resp;
+  }
   char val[32];
-  std::snprintf(val, sizeof val, "%g", series_value_.load());
+  std::snprintf(val, sizeof val, "%g", v);
   std::string result = series_json_zero_;
   // patch the placeholder value into every series
   std::string out;

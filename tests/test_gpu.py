@@ -233,3 +233,64 @@ def test_bench_one_gpu_quick():
     assert result["metric"] == "pods_evaluated_per_sec"
     assert result["value"] > 0
     assert "real GPU" in result["config"]["utilization_source"]
+
+
+def test_busy_gpu_is_not_culled_idle_gpu_is():
+    """Closed-loop semantic test on silicon: while the busy probe runs, the
+    idle query yields no candidates (the synthetic Prometheus honors the
+    == 0 predicate on the real activity value); once the GPU settles idle,
+    the same cluster is culled."""
+    _require_gpu()
+    import json
+    from gpu_pruner_amd import _gpumon, _pruner_core as core, probe
+
+    sampler = _gpumon.Sampler(poll_interval_ms=100)
+    sampler.init()
+    backend = core.SyntheticBackend(n_pods=20)
+    backend.start()
+    os.environ["GPU_PRUNER_K8S_URL"] = backend.k8s_url
+    os.environ["PROMETHEUS_TOKEN"] = "t"
+    cfg = json.dumps({"duration": 30, "grace_period": 300,
+                      "run_mode": "scale-down",
+                      "prometheus_url": backend.prom_url})
+    try:
+        # phase 1: GPU busy → no candidates, nothing scaled
+        probe.start(device=0, max_seconds=30.0)
+        try:
+            busy = 0.0
+            for _ in range(40):
+                time.sleep(0.1)
+                sampler.poll_once()
+                busy = sampler.snapshot()[0]["busy_percent"]
+                if busy >= 90.0:
+                    break
+            sampler.snapshot(True)
+            time.sleep(0.5)
+            sampler.poll_once()
+            ratio = sampler.snapshot()[0]["gr_engine_active"]
+            backend.set_series_value(ratio)
+            assert ratio > 0.0
+            out = core.run_tick(cfg)
+            assert out["num_unique_pods"] == 0, "busy GPU must yield no candidates"
+            assert backend.scale_patches == 0
+        finally:
+            probe.stop()
+
+        # phase 2: GPU idle → all pods candidates, parents culled
+        for _ in range(100):
+            time.sleep(0.1)
+            sampler.poll_once()
+            if sampler.snapshot()[0]["busy_percent"] == 0.0:
+                break
+        sampler.snapshot(True)
+        time.sleep(0.5)
+        sampler.poll_once()
+        ratio = sampler.snapshot()[0]["gr_engine_active"]
+        backend.set_series_value(ratio)
+        assert ratio == 0.0, f"idle ratio={ratio}"
+        out = core.run_tick(cfg)
+        assert out["num_unique_pods"] == 20
+        assert out["scaled"] == out["shutdown_events"] == 10
+    finally:
+        backend.stop()
+        sampler.stop()
