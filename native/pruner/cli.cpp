@@ -44,6 +44,9 @@ OPTIONS:
                                   failures [default: 5]
       --metrics-port <PORT>       serve daemon self-metrics + /healthz on
                                   this port (0 = disabled) [default: 0]
+      --eval-strategy <S>         candidate fetch strategy: get (per-object
+                                  GETs, reference-equivalent) | list
+                                  (namespace LISTs) | auto [default: auto]
   -h, --help                      print this help
 )";
 
@@ -83,6 +86,7 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
       {"--queue-capacity", "queue-capacity"},
       {"--max-failures", "max-failures"},
       {"--metrics-port", "metrics-port"},
+      {"--eval-strategy", "eval-strategy"},
       {"-h", "help"},              {"--help", "help"},
   };
   // flags that never take a value
@@ -171,6 +175,13 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
       else if (name == "queue-capacity") c.queue_capacity = std::stoi(val);
       else if (name == "max-failures") c.max_consecutive_failures = std::stoi(val);
       else if (name == "metrics-port") c.metrics_port = std::stoi(val);
+      else if (name == "eval-strategy") {
+        std::string m = strutil::lower(val);
+        if (m == "get") c.eval_strategy = EvalStrategy::PerPodGet;
+        else if (m == "list") c.eval_strategy = EvalStrategy::NamespaceList;
+        else if (m == "auto") c.eval_strategy = EvalStrategy::Auto;
+        else return fail("invalid --eval-strategy (expected get|list|auto): " + val);
+      }
     } catch (const std::exception&) {
       return fail("invalid value for --" + name + ": " + val);
     }

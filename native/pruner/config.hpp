@@ -11,6 +11,7 @@
 #include <string>
 #include <vector>
 
+#include "objcache.hpp"
 #include "promql.hpp"
 
 namespace pruner {
@@ -41,6 +42,11 @@ struct Config {
   int queue_capacity = 100;        // --queue-capacity
   int max_consecutive_failures = 5;  // --max-failures (abort after more than N)
   int metrics_port = 0;            // --metrics-port (0 = disabled)
+  // --eval-strategy: how candidate pods/owners are fetched —
+  //   get  = per-object GETs (reference-equivalent, 1-3 RTTs per pod)
+  //   list = namespace-collection LISTs (O(namespaces) RTTs per tick)
+  //   auto = LIST namespaces with >= 10 candidates, GETs elsewhere
+  EvalStrategy eval_strategy = EvalStrategy::Auto;
 
   QueryArgs query_args() const {
     QueryArgs qa;

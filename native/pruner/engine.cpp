@@ -18,6 +18,11 @@ constexpr const char* TARGET = "pruner::engine";
 }
 
 std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& pod) {
+  ObjectCache direct(kube, EvalStrategy::PerPodGet);
+  return find_root_object(direct, pod);
+}
+
+std::optional<ScaleKind> find_root_object(ObjectCache& objs, const jsn::Value& pod) {
   otlp::SpanGuard span("find_root_object");
   const jsn::Value& meta = pod.get("metadata");
   std::string pod_name = meta.get("name").as_string();
@@ -30,7 +35,7 @@ std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& po
   const jsn::Value& labels = meta.get("labels");
   const jsn::Value& ks = labels.get("serving.kserve.io/inferenceservice");
   if (ks.is_string()) {
-    auto is = kube.get_object(Kind::InferenceService, ns, ks.as_string());
+    auto is = objs.get_object(Kind::InferenceService, ns, ks.as_string());
     if (!is) {
       LOGW(TARGET, "KServe label points at missing InferenceService " + ks.as_string());
       return std::nullopt;
@@ -45,14 +50,14 @@ std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& po
       std::string owner_name = owner.get("name").as_string();
       if (owner_kind == "ReplicaSet") {
         LOGI(TARGET, "Found ReplicaSet!");
-        auto rs = kube.get_object(Kind::ReplicaSet, ns, owner_name);
+        auto rs = objs.get_object(Kind::ReplicaSet, ns, owner_name);
         if (rs) {
           const jsn::Value& rs_ors = rs->at({"metadata", "ownerReferences"});
           if (rs_ors.is_array()) {
             for (const auto& rs_or : rs_ors.arr()) {
               if (rs_or.get("kind").as_string() == "Deployment") {
                 LOGI(TARGET, "Found Deployment owning ReplicaSet!");
-                auto dep = kube.get_object(Kind::Deployment, ns, rs_or.get("name").as_string());
+                auto dep = objs.get_object(Kind::Deployment, ns, rs_or.get("name").as_string());
                 if (!dep) return std::nullopt;
                 return ScaleKind{Kind::Deployment, *dep};
               }
@@ -63,14 +68,14 @@ std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& po
         }
       } else if (owner_kind == "StatefulSet") {
         LOGI(TARGET, "Found StatefulSet!");
-        auto ss = kube.get_object(Kind::StatefulSet, ns, owner_name);
+        auto ss = objs.get_object(Kind::StatefulSet, ns, owner_name);
         if (ss) {
           const jsn::Value& ss_ors = ss->at({"metadata", "ownerReferences"});
           if (ss_ors.is_array()) {
             for (const auto& ss_or : ss_ors.arr()) {
               if (ss_or.get("kind").as_string() == "Notebook") {
                 LOGI(TARGET, "Found Notebook owning StatefulSet!");
-                auto nb = kube.get_object(Kind::Notebook, ns, ss_or.get("name").as_string());
+                auto nb = objs.get_object(Kind::Notebook, ns, ss_or.get("name").as_string());
                 if (!nb) return std::nullopt;
                 return ScaleKind{Kind::Notebook, *nb};
               }
@@ -180,6 +185,15 @@ std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& r
   LOGI(TARGET, "Query returned " + std::to_string(num_series) + " series across " +
                    std::to_string(unique_pods.size()) + " unique pods");
 
+  // LIST-vs-GET strategy: prefetch candidate namespaces' collections when
+  // it pays (see objcache.hpp).
+  ObjectCache objs(kube, cfg.eval_strategy);
+  {
+    std::map<std::string, int> ns_counts;
+    for (const auto& pmd : unique_pods) ns_counts[pmd.ns]++;
+    objs.prefetch(ns_counts, cfg.max_concurrency);
+  }
+
   // A pod must predate the whole lookback window (+ grace) for the "no
   // activity over the window" signal to be trustworthy.
   const double lookback_s =
@@ -195,7 +209,7 @@ std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& r
       unique_pods.size(), cfg.max_concurrency, [&](size_t i) {
       const PodMetricData& pmd = unique_pods[i];
       try {
-        auto pod = kube.get_pod(pmd.ns, pmd.name);
+        auto pod = objs.get_pod(pmd.ns, pmd.name);
         if (!pod) {
           LOGI(TARGET, "Skipping " + pmd.ns + ":" + pmd.name + ", pod no longer exists");
           return;
@@ -219,7 +233,7 @@ std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& r
         }
         if (created_s >= lookback_start) return;  // too young for the window
         LOGI(TARGET, "Pod " + pmd.ns + ":" + pmd.name + " is idle and eligible for scaledown");
-        results[i] = find_root_object(kube, *pod);
+        results[i] = find_root_object(objs, *pod);
       } catch (const std::exception& e) {
         LOGE(TARGET,
              "Skipping " + pmd.ns + ":" + pmd.name + ", retrieval error: " + e.what());

@@ -25,6 +25,7 @@
 #include "../common/queue.hpp"
 #include "config.hpp"
 #include "k8s.hpp"
+#include "objcache.hpp"
 #include "prom.hpp"
 #include "resources.hpp"
 
@@ -32,7 +33,9 @@ namespace pruner {
 
 // Walk the owner references of `pod` (a dynamic Pod object) to its scalable
 // root. Returns nullopt (with a log) when nothing scalable is found —
-// matching the reference's error path (lib.rs:509-512).
+// matching the reference's error path (lib.rs:509-512). The ObjectCache
+// overload answers from prefetched collections when available.
+std::optional<ScaleKind> find_root_object(ObjectCache& objs, const jsn::Value& pod);
 std::optional<ScaleKind> find_root_object(KubeClient& kube, const jsn::Value& pod);
 
 // Emit the scale Event (failure non-fatal) and apply the per-kind

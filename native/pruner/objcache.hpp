@@ -1,0 +1,61 @@
+// objcache.hpp — per-tick object source for the decision engine.
+//
+// The reference resolves each candidate with 1-3 apiserver GETs
+// (pod GET + owner walk), which bounds throughput at
+// RTT × calls-per-pod. This cache adds a LIST-based strategy: for
+// namespaces with many candidates, fetch the pod/owner collections once
+// (6 LISTs per namespace) and answer every lookup from memory — identical
+// decision semantics (a pod absent from a fresh LIST is exactly as gone as
+// a 404), two orders of magnitude fewer round-trips at 1000-pod scale.
+//
+//   get  — always per-object GETs (reference-equivalent behavior)
+//   list — always prefetch collections for candidate namespaces
+//   auto — LIST namespaces with >= threshold candidates, GETs elsewhere
+//          (big LISTs are not free on huge namespaces with few candidates)
+#pragma once
+
+#include <map>
+#include <optional>
+#include <set>
+#include <string>
+
+#include "k8s.hpp"
+#include "resources.hpp"
+
+namespace pruner {
+
+enum class EvalStrategy { PerPodGet, NamespaceList, Auto };
+
+class ObjectCache {
+public:
+  ObjectCache(KubeClient& kube, EvalStrategy strategy, int auto_threshold = 10)
+      : kube_(kube), strategy_(strategy), auto_threshold_(auto_threshold) {}
+
+  // Decide which namespaces to prefetch given candidate counts, and fetch
+  // their collections concurrently (pods + the five scalable kinds).
+  // `concurrency` bounds the parallel LIST fan-out.
+  void prefetch(const std::map<std::string, int>& ns_candidate_counts, int concurrency);
+
+  // Lookups mirror KubeClient semantics (nullopt = not found).
+  std::optional<jsn::Value> get_pod(const std::string& ns, const std::string& name);
+  std::optional<jsn::Value> get_object(Kind kind, const std::string& ns,
+                                       const std::string& name);
+
+  KubeClient& kube() { return kube_; }
+  size_t lists_issued() const { return lists_issued_; }
+
+private:
+  struct NsCache {
+    // kind ("Pod" or kind_name) → name → object; a kind appears only if its
+    // LIST succeeded (CRDs may be absent from a cluster → fall back to GET)
+    std::map<std::string, std::map<std::string, jsn::Value>> by_kind;
+  };
+
+  KubeClient& kube_;
+  EvalStrategy strategy_;
+  int auto_threshold_;
+  std::map<std::string, NsCache> cache_;  // namespace → cache
+  size_t lists_issued_ = 0;
+};
+
+}  // namespace pruner

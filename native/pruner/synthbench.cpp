@@ -257,6 +257,25 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
 
   std::lock_guard<std::mutex> lock(mu_);
   auto kit = objects_.find(kind);
+  // namespaced collection LIST (no object name)
+  if (req.method == "GET" && name.empty()) {
+    std::string body = "{\"kind\":\"" + kind + "List\",\"items\":[";
+    bool first = true;
+    if (kit != objects_.end()) {
+      auto nit = kit->second.find(ns);
+      if (nit != kit->second.end()) {
+        for (auto& [oname, stored] : nit->second) {
+          if (!first) body += ",";
+          first = false;
+          if (stored.cached_dump.empty()) stored.cached_dump = stored.obj.dump();
+          body += stored.cached_dump;
+        }
+      }
+    }
+    body += "]}";
+    resp.body = std::move(body);
+    return resp;
+  }
   StoredObject* obj = nullptr;
   if (kit != objects_.end()) {
     auto nit = kit->second.find(ns);

@@ -61,6 +61,10 @@ Config config_from_json(const std::string& json_cfg) {
   if (v.get("run_mode").as_string_or("dry-run") == "scale-down")
     c.run_mode = RunMode::ScaleDown;
   c.enabled_resources = v.get("enabled_resources").as_string_or("drsin");
+  std::string strat = v.get("eval_strategy").as_string_or("auto");
+  c.eval_strategy = strat == "get"    ? EvalStrategy::PerPodGet
+                    : strat == "list" ? EvalStrategy::NamespaceList
+                                      : EvalStrategy::Auto;
   if (v.get("prometheus_url").is_string())
     c.prometheus_url = v.get("prometheus_url").as_string();
   return c;

@@ -250,3 +250,50 @@ def test_repeat_tick_idempotent(core, fake_api):
     for _ in range(3):
         out = evaluate(core, result)
         assert out["shutdown_events"] == 1
+
+
+@pytest.mark.parametrize("strategy", ["get", "list", "auto"])
+def test_eval_strategies_identical_outcomes(core, fake_api, strategy):
+    """LIST-based evaluation must reach exactly the per-GET decisions
+    (native/pruner/objcache.hpp)."""
+    # mixed cluster incl. a gone pod, a pending pod, and a KServe shortcut
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    for i in range(12):
+        fake_api.add_pod(f"p{i}", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                         owner_uid=rs["metadata"]["uid"])
+    nb = fake_api.add_notebook("wb", "ml")
+    fake_api.add_statefulset("wb-ss", "ml", notebook_owner=nb)
+    fake_api.add_pod("wb-ss-0", "ml", owner_kind="StatefulSet", owner_name="wb-ss")
+    fake_api.add_inferenceservice("llm", "ml")
+    fake_api.add_pod("pred-0", "ml",
+                     labels={"serving.kserve.io/inferenceservice": "llm"})
+    fake_api.add_pod("pending", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], phase="Pending")
+    result = ([series(f"p{i}", "ml") for i in range(12)] +
+              [series("wb-ss-0", "ml"), series("pred-0", "ml"),
+               series("pending", "ml"), series("ghost", "ml")])
+    cfg = dict(CFG, eval_strategy=strategy)
+    out = evaluate(core, result, cfg)
+    assert out["num_unique_pods"] == 16
+    assert out["shutdown_events"] == 3  # Deployment + Notebook + InferenceService
+    assert sorted(r.kind for r in out["roots"]) == [
+        "Deployment", "InferenceService", "Notebook"]
+
+
+def test_list_strategy_uses_few_requests(core, fake_api):
+    """At 40 candidates in one namespace the LIST path costs O(1) requests."""
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    for i in range(40):
+        fake_api.add_pod(f"p{i}", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                         owner_uid=rs["metadata"]["uid"])
+    result = [series(f"p{i}", "ml") for i in range(40)]
+    fake_api.requests.clear()
+    evaluate(core, result, dict(CFG, eval_strategy="list"))
+    n_list = len(fake_api.requests)
+    fake_api.requests.clear()
+    evaluate(core, result, dict(CFG, eval_strategy="get"))
+    n_get = len(fake_api.requests)
+    assert n_list <= 8, f"LIST strategy made {n_list} requests"
+    assert n_get >= 40, f"GET strategy made {n_get} requests"
