@@ -57,6 +57,9 @@ def main():
     ap.add_argument("--concurrency", type=int,
                     default=int(os.environ.get("BENCH_CONCURRENCY", "32")),
                     help="engine --max-concurrency")
+    ap.add_argument("--otlp", action="store_true",
+                    help="export spans/metrics to an in-process OTLP collector "
+                         "during the timed region (BASELINE config 5)")
     args = ap.parse_args()
 
     import torch
@@ -89,6 +92,15 @@ def main():
     backend.start()
     os.environ["GPU_PRUNER_K8S_URL"] = backend.k8s_url
     os.environ["PROMETHEUS_TOKEN"] = "bench-token"
+
+    otlp_collector = None
+    if args.otlp:
+        from gpu_pruner_amd.fixtures import FakeOtlpCollector
+
+        otlp_collector = FakeOtlpCollector().start()
+        os.environ["OTEL_EXPORTER_OTLP_ENDPOINT"] = otlp_collector.url
+        os.environ["OTEL_METRIC_EXPORT_INTERVAL"] = "1000"
+        core.otlp_init("gpu-pruner-bench")
     cfg = json.dumps({
         "duration": 30, "grace_period": 300, "run_mode": "scale-down",
         "prometheus_url": backend.prom_url, "max_concurrency": args.concurrency,
@@ -168,6 +180,11 @@ def main():
     backend.stop()
     if sampler is not None:
         sampler.stop()
+    otlp_spans = None
+    if otlp_collector is not None:
+        core.otlp_shutdown()
+        otlp_spans = len(otlp_collector.span_names())
+        otlp_collector.stop()
 
     if rank == 0:
         result = {
@@ -195,6 +212,8 @@ def main():
                 "events_posted": backend.events_posted,
                 "utilization_source": "rocm_smi sampler (real GPU)" if sampler else
                                       "synthetic idle (no GPU)",
+                "otlp": ("enabled, %d spans exported" % otlp_spans)
+                        if otlp_spans is not None else "disabled",
             },
         }
         print(json.dumps(result), flush=True)

@@ -18,6 +18,7 @@
 #include "../pruner/promql.hpp"
 #include "../pruner/resources.hpp"
 #include "../common/http.hpp"
+#include "../pruner/otlp.hpp"
 #include "../pruner/synthbench.hpp"
 
 namespace py = pybind11;
@@ -279,6 +280,14 @@ PYBIND11_MODULE(_pruner_core, m) {
       .def_property_readonly("scale_patches", &SyntheticBackend::scale_patches)
       .def_property_readonly("requests_served", &SyntheticBackend::requests_served)
       .def_property_readonly("expected_parents", &SyntheticBackend::expected_parents);
+
+  // OTLP lifecycle (the binaries call these in main(); bench.py/config-5
+  // runs need them from Python)
+  m.def("otlp_init", [](const std::string& service) { otlp::init(service); },
+        py::arg("service") = "gpu-pruner");
+  m.def("otlp_shutdown", [] { otlp::shutdown(); },
+        py::call_guard<py::gil_scoped_release>());
+  m.def("otlp_enabled", [] { return otlp::enabled(); });
 
   // test helper: raw GET through the native HTTP client (used to pin
   // chunked / close-delimited / keep-alive decoding against fixture servers)
