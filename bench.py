@@ -137,7 +137,6 @@ def main():
     # ---- warmup ----
     for _ in range(args.warmup):
         out = one_step()
-    expected = {"num_unique_pods": pods_per_rank}
     if out["num_unique_pods"] != pods_per_rank:
         log(f"[bench] WARNING rank {rank}: evaluated {out['num_unique_pods']} != "
             f"{pods_per_rank} pods (GPU busy? series value nonzero)")
