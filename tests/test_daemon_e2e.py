@@ -193,3 +193,17 @@ def test_multi_pod_mixed_cluster(pruner_bin, fake_api, fake_prom):
     assert scaled == info["expected_shutdown_events"]
     # every action announced
     assert len(fake_api.events) == info["expected_shutdown_events"]
+
+
+def test_scale_down_with_list_strategy(pruner_bin, fake_api, fake_prom):
+    """The daemon binary with --eval-strategy list reaches identical outcomes."""
+    from gpu_pruner_amd.fixtures import build_synthetic_cluster
+
+    info = build_synthetic_cluster(fake_api, fake_prom, n_pods=30, pods_per_parent=3)
+    r = run_pruner(pruner_bin, fake_api, fake_prom, "--run-mode", "scale-down",
+                   "--eval-strategy", "list")
+    assert r.returncode == 0, r.stderr
+    assert len(fake_api.events) == info["expected_shutdown_events"]
+    # LIST path: far fewer GETs than pods
+    gets = [p for (m, p) in fake_api.requests if m == "GET" and "/pods/" in p]
+    assert len(gets) == 0, f"list strategy should not GET individual pods: {gets[:3]}"
