@@ -272,8 +272,8 @@ private:
 
 class ParseError : public std::runtime_error {
 public:
-  ParseError(const std::string& msg, size_t pos)
-      : std::runtime_error(msg + " at offset " + std::to_string(pos)), pos(pos) {}
+  ParseError(const std::string& msg, size_t at)
+      : std::runtime_error(msg + " at offset " + std::to_string(at)), pos(at) {}
   size_t pos;
 };
 

@@ -39,8 +39,8 @@ struct KubeConfig {
 
 class KubeError : public std::runtime_error {
 public:
-  KubeError(int status, const std::string& msg)
-      : std::runtime_error(msg), status(status) {}
+  KubeError(int http_status, const std::string& msg)
+      : std::runtime_error(msg), status(http_status) {}
   int status;  // HTTP status; 0 for transport errors
 };
 
