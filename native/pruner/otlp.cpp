@@ -22,8 +22,12 @@ struct FinishedSpan {
   uint64_t end_ns;
 };
 
+constexpr size_t kMaxBufferedSpans = 50000;  // drop-oldest beyond this
+constexpr size_t kMaxSpansPerPost = 5000;    // bound per-request payload
+
 struct State {
   std::atomic<bool> enabled{false};
+  std::atomic<uint64_t> dropped_spans{0};
   std::atomic<bool> running{false};
   std::atomic<uint64_t> delivered{0};
   std::string endpoint;  // base, no trailing slash
@@ -82,13 +86,16 @@ void post_json(const std::string& url, const jsn::Value& body) {
 
 void export_once() {
   State& s = state();
-  // ---- spans ----
-  std::vector<FinishedSpan> batch;
+  // ---- spans (chunked: a slow collector must not grow our heap) ----
+  std::vector<FinishedSpan> all;
   {
     std::lock_guard<std::mutex> lock(s.mu);
-    batch.swap(s.spans);
+    all.swap(s.spans);
   }
-  if (!batch.empty()) {
+  for (size_t base = 0; base < all.size(); base += kMaxSpansPerPost) {
+    size_t n = std::min(kMaxSpansPerPost, all.size() - base);
+    std::vector<FinishedSpan> batch(all.begin() + static_cast<long>(base),
+                                    all.begin() + static_cast<long>(base + n));
     jsn::Value spans = jsn::Value::array();
     for (const auto& fs : batch) {
       jsn::Value sp = jsn::Value::object();
@@ -205,6 +212,13 @@ SpanGuard::~SpanGuard() {
   if (!enabled()) return;
   State& s = state();
   std::lock_guard<std::mutex> lock(s.mu);
+  if (s.spans.size() >= kMaxBufferedSpans) {
+    // bounded buffer: a stalled exporter drops oldest spans instead of
+    // growing the daemon heap without limit
+    s.spans.erase(s.spans.begin(),
+                  s.spans.begin() + static_cast<long>(kMaxBufferedSpans / 10));
+    s.dropped_spans.fetch_add(kMaxBufferedSpans / 10, std::memory_order_relaxed);
+  }
   s.spans.push_back({name_, start_ns_, now_unix_ns()});
 }
 
