@@ -36,6 +36,9 @@ REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)
 
 os.environ.setdefault("GPU_PRUNER_LOG", "error")
+# cap glibc malloc arenas: the engine's 256-thread pool otherwise grows RSS
+# toward N_arenas x high-water (see native/pruner/main.cpp)
+os.environ.setdefault("MALLOC_ARENA_MAX", "2")
 
 TOTAL_PODS_DEFAULT = 1000
 

