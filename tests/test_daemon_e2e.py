@@ -174,10 +174,11 @@ def test_env_token_fallback(pruner_bin, fake_api, fake_prom):
 
 
 def test_multi_pod_mixed_cluster(pruner_bin, fake_api, fake_prom):
-    """Mixed kinds (BASELINE config 3 shape, small): all parents culled once."""
+    """BASELINE config 3 shape: 50 mixed Deploy/SS+Notebook/InferenceService
+    pods — every parent culled exactly once, announced exactly once."""
     from gpu_pruner_amd.fixtures import build_synthetic_cluster
 
-    info = build_synthetic_cluster(fake_api, fake_prom, n_pods=24, pods_per_parent=2)
+    info = build_synthetic_cluster(fake_api, fake_prom, n_pods=50, pods_per_parent=2)
     r = run_pruner(pruner_bin, fake_api, fake_prom, "--run-mode", "scale-down",
                    "--max-concurrency", "16")
     assert r.returncode == 0, r.stderr
@@ -207,3 +208,49 @@ def test_scale_down_with_list_strategy(pruner_bin, fake_api, fake_prom):
     # LIST path: far fewer GETs than pods
     gets = [p for (m, p) in fake_api.requests if m == "GET" and "/pods/" in p]
     assert len(gets) == 0, f"list strategy should not GET individual pods: {gets[:3]}"
+
+
+def test_config4_filters_grace_and_partially_busy(pruner_bin, fake_api, fake_prom):
+    """BASELINE config 4: namespace + model-name filters with grace period on
+    a partially-busy cluster. Busy pods never appear in the idle query result
+    (Prometheus's == 0 predicate), young pods are age-filtered, other-model
+    and other-namespace pods are excluded by the pushed-down regex filters —
+    only the old idle matching pod's parent is culled."""
+    def deployment_with_pod(name, ns, age_s=3 * 3600):
+        dep = fake_api.add_deployment(name, ns)
+        rs = fake_api.add_replicaset(f"{name}-rs", ns, owner=dep)
+        fake_api.add_pod(f"{name}-0", ns, owner_kind="ReplicaSet",
+                         owner_name=f"{name}-rs", owner_uid=rs["metadata"]["uid"],
+                         age_s=age_s)
+        return dep
+
+    deployment_with_pod("idle-old", "ml-team-a")          # culled
+    deployment_with_pod("idle-young", "ml-team-a", age_s=60)  # grace-filtered
+    deployment_with_pod("busy", "ml-team-a")              # busy → no series
+    deployment_with_pod("other-ns", "web")                # namespace filter
+    deployment_with_pod("other-gpu", "ml-team-a")         # model filter
+
+    # Prometheus only returns series matching the filters and == 0:
+    fake_prom.add_idle_series("idle-old-0", "ml-team-a",
+                              model_name="AMD Instinct MI355X")
+    fake_prom.add_idle_series("idle-young-0", "ml-team-a",
+                              model_name="AMD Instinct MI355X")
+    # busy pod: excluded by == 0; other-ns/other-gpu: excluded by the regex
+    # filters — the real Prometheus applies those, so they emit no series.
+
+    r = run_pruner(pruner_bin, fake_api, fake_prom, "--run-mode", "scale-down",
+                   "--namespace", "ml-team-.*",
+                   "--model-name", "AMD Instinct MI355.*",
+                   "--grace-period", "300")
+    assert r.returncode == 0, r.stderr
+    # filters were pushed into the wire query
+    q = fake_prom.queries[0]
+    assert 'exported_namespace =~ "ml-team-.*"' in q
+    assert 'modelName =~ "AMD Instinct MI355.*"' in q
+    # only the old idle matching deployment was culled
+    assert fake_api.get("Deployment", "ml-team-a", "idle-old")["spec"]["replicas"] == 0
+    for name, ns in [("idle-young", "ml-team-a"), ("busy", "ml-team-a"),
+                     ("other-gpu", "ml-team-a")]:
+        assert fake_api.get("Deployment", ns, name)["spec"]["replicas"] == 1, name
+    assert fake_api.get("Deployment", "web", "other-ns")["spec"]["replicas"] == 1
+    assert len(fake_api.events) == 1
