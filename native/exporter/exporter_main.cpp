@@ -90,6 +90,9 @@ int main(int argc, char** argv) {
       fmt = v == "json" ? logx::Format::Json
             : v == "pretty" ? logx::Format::Pretty
                             : logx::Format::Default;
+    } else if (a == "--version" || a == "-V") {
+      std::puts("mi355-exporter 0.1.0 (MI355X-native)");
+      return 0;
     } else if (a == "-h" || a == "--help") {
       std::fputs(HELP, stdout);
       return 0;

@@ -27,6 +27,10 @@ void cap_malloc_arenas() {
 int main(int argc, char** argv) {
   cap_malloc_arenas();
   std::vector<std::string> args(argv + 1, argv + argc);
+  if (args.size() == 1 && (args[0] == "--version" || args[0] == "-V")) {
+    std::puts("gpu-pruner 0.1.0 (MI355X-native)");
+    return 0;
+  }
   pruner::CliResult cli = pruner::parse_cli(args);
   if (cli.show_help) {
     std::fputs(pruner::cli_help().c_str(), stdout);
