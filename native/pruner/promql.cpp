@@ -6,6 +6,20 @@ namespace pruner {
 
 namespace {
 
+// Escape a value for embedding inside a double-quoted PromQL string
+// (regex matchers): backslashes and quotes. The reference interpolates the
+// raw flag value into its template (query.promql.j2:12), so a quote in
+// --namespace breaks its query; here it cannot.
+std::string esc(const std::string& s) {
+  std::string out;
+  out.reserve(s.size());
+  for (char c : s) {
+    if (c == '\\' || c == '"') out += '\\';
+    out += c;
+  }
+  return out;
+}
+
 std::string fmt_num(double v) {
   char buf[64];
   // integral thresholds render without a decimal point ("150"), matching the
@@ -30,8 +44,9 @@ std::string build_idle_query(const QueryArgs& a) {
   // the optional namespace / model regex filters.
   auto selector = [&](bool with_model) {
     std::string s = "{\n      " + pl + " != \"\"";
-    if (a.namespace_re) s += ", " + nl + " =~ \"" + *a.namespace_re + "\"";
-    if (with_model && a.model_name_re) s += ", modelName =~ \"" + *a.model_name_re + "\"";
+    if (a.namespace_re) s += ", " + nl + " =~ \"" + esc(*a.namespace_re) + "\"";
+    if (with_model && a.model_name_re)
+      s += ", modelName =~ \"" + esc(*a.model_name_re) + "\"";
     s += "\n    }";
     return s;
   };

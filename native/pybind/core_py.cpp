@@ -9,6 +9,8 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <chrono>
+
 #include "../common/json.hpp"
 #include "../common/log.hpp"
 #include "../pruner/config.hpp"
@@ -218,23 +220,35 @@ PYBIND11_MODULE(_pruner_core, m) {
   m.def("run_tick",
         [](const std::string& cfg_json) {
           Config cfg = config_from_json(cfg_json);
+          auto t0 = std::chrono::steady_clock::now();
           auto prom = build_prom_client(cfg);
           KubeClient kube(KubeConfig::resolve());
           std::string query = build_idle_query(cfg.query_args());
           jsn::Value result = prom->query_vector(query);
+          auto t1 = std::chrono::steady_clock::now();
           QueryOutcome out;
           std::vector<ScaleKind> roots = evaluate_candidates(kube, result, cfg, &out);
+          auto t2 = std::chrono::steady_clock::now();
           size_t scaled = 0;
           if (cfg.run_mode == RunMode::ScaleDown) {
             uint8_t enabled = get_enabled_resources(cfg.enabled_resources);
             scaled = scale_all(kube, roots, enabled, cfg.max_concurrency);
           }
+          auto t3 = std::chrono::steady_clock::now();
+          auto ms = [](auto a, auto b) {
+            return std::chrono::duration<double, std::milli>(b - a).count();
+          };
           py::gil_scoped_acquire gil;
           py::dict d;
           d["num_series"] = out.num_series;
           d["num_unique_pods"] = out.num_unique_pods;
           d["shutdown_events"] = out.shutdown_events;
           d["scaled"] = scaled;
+          py::dict phases;
+          phases["query_ms"] = ms(t0, t1);
+          phases["evaluate_ms"] = ms(t1, t2);
+          phases["actuate_ms"] = ms(t2, t3);
+          d["phase_ms"] = phases;
           return d;
         },
         py::call_guard<py::gil_scoped_release>(),

@@ -104,3 +104,14 @@ def test_query_power_threshold_fractional(core):
 def test_query_duration_range(core, duration):
     q = render(core, {"duration": duration})
     assert f"[{duration}m]" in q
+
+
+def test_query_escapes_quotes_in_filters(core):
+    """A quote in --namespace/--model-name must not break out of the PromQL
+    string matcher (the reference interpolates the raw value)."""
+    q = render(core, {"duration": 30, "namespace": 'evil"} or up{x="',
+                      "model_name": 'A"B\\C'})
+    assert 'evil\\"} or up{x=\\"' in q
+    assert 'A\\"B\\\\C' in q
+    # the query still terminates with the idle predicate
+    assert q.rstrip().endswith("== 0")
