@@ -4,7 +4,6 @@
 
 #include <atomic>
 #include <chrono>
-#include <condition_variable>
 #include <memory>
 #include <mutex>
 #include <sstream>
@@ -23,16 +22,23 @@ namespace pruner {
 namespace {
 constexpr const char* TARGET = "pruner::daemon";
 
-// Graceful shutdown: SIGTERM/SIGINT set the flag; the interval wait is a
-// condition-variable sleep so K8s pod termination interrupts a 180 s tick
-// wait immediately, finishes in-flight scale actions, and exits 0.
+// Graceful shutdown: SIGTERM/SIGINT set the flag (only an atomic store —
+// the async-signal-safe subset); the interval wait polls it in 100 ms
+// slices so K8s pod termination interrupts a 180 s tick wait promptly,
+// finishes in-flight scale actions, and exits 0.
 std::atomic<bool> g_shutdown{false};
-std::mutex g_shutdown_mu;
-std::condition_variable g_shutdown_cv;
 
-void on_shutdown_signal(int) {
-  g_shutdown.store(true);
-  g_shutdown_cv.notify_all();
+void on_shutdown_signal(int) { g_shutdown.store(true); }
+
+// Sleep until `deadline` or shutdown, whichever first.
+void interruptible_sleep_until(std::chrono::steady_clock::time_point deadline) {
+  while (!g_shutdown.load(std::memory_order_relaxed)) {
+    auto now = std::chrono::steady_clock::now();
+    if (now >= deadline) return;
+    auto slice = std::min<std::chrono::steady_clock::duration>(
+        deadline - now, std::chrono::milliseconds(100));
+    std::this_thread::sleep_for(slice);
+  }
 }
 
 // Daemon self-metrics in Prometheus text format (--metrics-port): the six
@@ -133,8 +139,7 @@ int run_daemon(const Config& cfg) {
   auto next_tick = std::chrono::steady_clock::now();
   while (!g_shutdown.load()) {
     if (cfg.daemon_mode) {
-      std::unique_lock<std::mutex> lock(g_shutdown_mu);
-      g_shutdown_cv.wait_until(lock, next_tick, [] { return g_shutdown.load(); });
+      interruptible_sleep_until(next_tick);
       if (g_shutdown.load()) break;
       next_tick += std::chrono::seconds(cfg.check_interval_s);
     }

@@ -4,6 +4,7 @@
 #include <chrono>
 #include <condition_variable>
 #include <mutex>
+#include <memory>
 #include <thread>
 #include <vector>
 
@@ -39,6 +40,7 @@ struct State {
   std::condition_variable cv;
   bool stop = false;
   int interval_ms = 5000;
+  std::unique_ptr<http::Client> client;  // persistent export connection pool
 };
 
 State& state() {
@@ -68,20 +70,23 @@ jsn::Value resource_json(const std::string& service_name) {
 }
 
 void post_json(const std::string& url, const jsn::Value& body) {
+  State& s = state();
   auto parsed = http::Url::parse(url);
   if (!parsed) return;
-  http::ClientOptions opts;
-  opts.connect_timeout_ms = 2000;
-  opts.io_timeout_ms = 5000;
-  http::Client client(*parsed, opts);
+  if (!s.client) {
+    http::ClientOptions opts;
+    opts.connect_timeout_ms = 2000;
+    opts.io_timeout_ms = 5000;
+    s.client = std::make_unique<http::Client>(*parsed, opts);
+  }
   http::Request req;
   req.method = "POST";
   req.path = parsed->path;
   req.body = body.dump();
   req.headers.emplace_back("Content-Type", "application/json");
-  http::Response resp = client.request(req);
+  http::Response resp = s.client->request(req);
   if (resp.status >= 200 && resp.status < 300)
-    state().delivered.fetch_add(1, std::memory_order_relaxed);
+    s.delivered.fetch_add(1, std::memory_order_relaxed);
 }
 
 void export_once() {
