@@ -49,3 +49,16 @@ def test_bench_single_rank_json_contract():
         assert key in result, f"missing contract key {key}"
     assert result["data"] == "synthetic"
     assert result["config"]["p50_scale_decision_latency_ms"] > 0
+
+
+def test_bench_latency_injection_flag():
+    """--latency-us feeds through to the synthetic apiserver."""
+    r = subprocess.run(
+        [sys.executable, str(REPO_ROOT / "bench.py"), "--steps", "2",
+         "--warmup", "1", "--pods", "40", "--latency-us", "2000"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO_ROOT))
+    assert r.returncode == 0, r.stderr[-2000:]
+    result = json.loads(r.stdout.strip().splitlines()[-1])
+    assert result["config"]["apiserver_latency_us"] == 2000
+    # with 2ms per request the tick cannot be instant
+    assert result["ms_per_step"] > 4

@@ -254,3 +254,13 @@ def test_config4_filters_grace_and_partially_busy(pruner_bin, fake_api, fake_pro
         assert fake_api.get("Deployment", ns, name)["spec"]["replicas"] == 1, name
     assert fake_api.get("Deployment", "web", "other-ns")["spec"]["replicas"] == 1
     assert len(fake_api.events) == 1
+
+
+def test_non_vector_response_counts_as_failure(pruner_bin, fake_api, fake_prom):
+    """A matrix (range) response is a query failure, not a crash — the
+    breaker handles it (reference expects a vector, main.rs:405-409)."""
+    fake_prom.data_override = {"resultType": "matrix", "result": []}
+    r = run_pruner(pruner_bin, fake_api, fake_prom)
+    # one-shot: failure logged, exit 0 (breaker only trips in daemon mode)
+    assert r.returncode == 0
+    assert "expected vector" in r.stderr
