@@ -368,6 +368,7 @@ Response Client::request(const Request& req) {
 }
 
 Response Client::do_request_on(Connection& c, const Request& req) {
+  // (interim 1xx responses are skipped below without re-sending the request)
   std::string host_hdr =
       base_.scheme == "unix" ? "localhost" : base_.host + ":" + std::to_string(base_.port);
   std::string out;
@@ -397,9 +398,11 @@ Response Client::do_request_on(Connection& c, const Request& req) {
 
   c.write_all(out.data(), out.size());
 
-  // ---- read status line + headers ----
+  // ---- read status line + headers (skipping any interim 1xx) ----
   std::string buf;
-  auto read_until_headers = [&]() {
+  Response resp;
+  std::string rest;
+  while (true) {
     char tmp[8192];
     while (buf.find("\r\n\r\n") == std::string::npos) {
       size_t r = c.read_some(tmp, sizeof tmp);
@@ -413,34 +416,33 @@ Response Client::do_request_on(Connection& c, const Request& req) {
         throw Error("response headers too large");
       }
     }
-  };
-  read_until_headers();
-
-  Response resp;
-  size_t hdr_end = buf.find("\r\n\r\n");
-  std::string head = buf.substr(0, hdr_end);
-  std::string rest = buf.substr(hdr_end + 4);
-  auto lines = strutil::split(head, '\n');
-  if (lines.empty()) throw Error("malformed response");
-  {
-    std::string status_line = strutil::trim(lines[0]);
-    // HTTP/1.1 200 OK
-    size_t sp1 = status_line.find(' ');
-    if (sp1 == std::string::npos) throw Error("malformed status line: " + status_line);
-    resp.status = std::atoi(status_line.c_str() + sp1 + 1);
-  }
-  for (size_t i = 1; i < lines.size(); i++) {
-    std::string line = strutil::trim(lines[i]);
-    size_t colon = line.find(':');
-    if (colon == std::string::npos) continue;
-    resp.headers[strutil::lower(line.substr(0, colon))] = strutil::trim(line.substr(colon + 1));
-  }
-
-  // 100-continue: skip interim responses
-  if (resp.status == 100) {
-    buf = rest;
-    read_until_headers();
-    return do_request_on(c, req);  // unlikely path; simplest correct handling
+    resp = Response{};
+    size_t hdr_end = buf.find("\r\n\r\n");
+    std::string head = buf.substr(0, hdr_end);
+    rest = buf.substr(hdr_end + 4);
+    auto lines = strutil::split(head, '\n');
+    if (lines.empty()) throw Error("malformed response");
+    {
+      std::string status_line = strutil::trim(lines[0]);
+      // HTTP/1.1 200 OK
+      size_t sp1 = status_line.find(' ');
+      if (sp1 == std::string::npos) throw Error("malformed status line: " + status_line);
+      resp.status = std::atoi(status_line.c_str() + sp1 + 1);
+    }
+    for (size_t i = 1; i < lines.size(); i++) {
+      std::string line = strutil::trim(lines[i]);
+      size_t colon = line.find(':');
+      if (colon == std::string::npos) continue;
+      resp.headers[strutil::lower(line.substr(0, colon))] =
+          strutil::trim(line.substr(colon + 1));
+    }
+    if (resp.status >= 100 && resp.status < 200) {
+      // interim response (100-continue etc.): it has no body; keep reading
+      // for the real one without re-sending anything
+      buf = rest;
+      continue;
+    }
+    break;
   }
 
   bool keep_alive = true;

@@ -99,3 +99,14 @@ def test_status_and_headers_parsed(core, raw_server):
     assert r["status"] == 404
     assert r["headers"]["x-custom"] == "Value"
     assert r["body"] == b"not found"
+
+
+def test_interim_100_continue_skipped(core, raw_server):
+    """An interim 100 response is skipped without re-sending the request."""
+    resp = (b"HTTP/1.1 100 Continue\r\n\r\n"
+            b"HTTP/1.1 200 OK\r\nContent-Length: 4\r\n\r\ndone")
+    url, state = raw_server([resp])
+    r = core._http_get(url + "/x")
+    assert r["status"] == 200
+    assert r["body"] == b"done"
+    assert len(state["requests"]) == 1  # no duplicate request
