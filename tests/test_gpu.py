@@ -144,6 +144,12 @@ def test_exporter_binary_serves_real_metrics():
         assert "DCGM_FI_DEV_POWER_USAGE{" in text
         assert 'node_type="amd-mi355x"' in text
         assert 'modelName=' in text
+        # full family surface incl. the AMD-native xGMI extras
+        for fam in ("DCGM_FI_DEV_FB_USED", "DCGM_FI_DEV_FB_FREE",
+                    "DCGM_FI_DEV_GPU_TEMP", "DCGM_FI_DEV_SM_CLOCK",
+                    "DCGM_FI_DEV_TOTAL_ENERGY_CONSUMPTION",
+                    "mi355_xgmi_link_width", "mi355_xgmi_read_kb_total"):
+            assert f"# TYPE {fam} " in text, f"missing family {fam}" 
         health = urllib.request.urlopen(
             f"http://127.0.0.1:{port}/healthz", timeout=2).read()
         assert health == b"ok\n"
