@@ -123,3 +123,27 @@ def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
     finally:
         p.terminate()
         p.wait(timeout=10)
+
+
+def test_five_thousand_pod_tick_correctness(core, monkeypatch):
+    """10x-BASELINE stress (native backend): one tick evaluates 5000 pods and
+    culls every parent exactly once — guards against O(n^2) regressions and
+    dedup breakage at scale. (Timing is profiled separately; this asserts
+    correctness only.)"""
+    import json as _json
+
+    monkeypatch.setenv("PROMETHEUS_TOKEN", "t")
+    b = core.SyntheticBackend(n_pods=5000)
+    b.start()
+    try:
+        monkeypatch.setenv("GPU_PRUNER_K8S_URL", b.k8s_url)
+        cfg = _json.dumps({"duration": 30, "grace_period": 300,
+                           "run_mode": "scale-down",
+                           "prometheus_url": b.prom_url, "max_concurrency": 32})
+        out = core.run_tick(cfg)
+        assert out["num_unique_pods"] == 5000
+        assert out["shutdown_events"] == 2500
+        assert out["scaled"] == 2500
+        assert b.events_posted == 2500
+    finally:
+        b.stop()
