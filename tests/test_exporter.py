@@ -166,3 +166,22 @@ def test_attribution_unowned_gpu(gpumon, fake_node, monkeypatch):
     a = gpumon.Attributor()
     # kfd id 888 has no pids → no attribution
     assert a.resolve([(0, 888)]) == {}
+
+
+def test_exporter_binary_fails_loudly_without_gpu():
+    """On a host without amdgpu the exporter must exit nonzero with a clear
+    error — never serve fake zeros (the == 0 predicate would cull everything)."""
+    import subprocess
+    from pathlib import Path
+
+    binary = Path(__file__).resolve().parent.parent / "bin" / "mi355-exporter"
+    if not binary.exists():
+        pytest.skip("exporter not built")
+    import torch
+
+    if torch.cuda.is_available():
+        pytest.skip("host has a GPU; negative test is for CPU CI")
+    r = subprocess.run([str(binary), "-p", "19499"], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode == 1
+    assert "sampler init failed" in r.stderr
