@@ -39,7 +39,8 @@ cfg = json.dumps({"duration": 30, "grace_period": 300, "run_mode": "scale-down",
                   "prometheus_url": backend.prom_url})
 
 phases = []   # (phase, ticks, scaled_total)
-t_end = time.time() + 60
+import os as _os
+t_end = time.time() + int(_os.environ.get("SOAK_SECONDS", "60"))
 phase_idx = 0
 ok = True
 try:
