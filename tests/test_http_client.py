@@ -110,3 +110,40 @@ def test_interim_100_continue_skipped(core, raw_server):
     assert r["status"] == 200
     assert r["body"] == b"done"
     assert len(state["requests"]) == 1  # no duplicate request
+
+
+def test_server_survives_malformed_requests(core):
+    """The exporter's HTTP server is network-exposed in-cluster: garbage,
+    partial, and oversized requests must not crash or wedge it."""
+    import socket
+
+    b = core.SyntheticBackend(n_pods=2)
+    b.start()
+    try:
+        host, port = b.prom_url.rsplit(":", 1)[0].split("//")[1], int(b.prom_url.rsplit(":", 1)[1])
+        payloads = [
+            b"\x00\x01\x02\xff\xfe garbage\r\n\r\n",
+            b"GET\r\n\r\n",                       # missing path
+            b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n",  # h2 preface at an h1 server
+            b"GET / HTTP/1.1\r\nContent-Length: 99999999\r\n\r\nshort",
+            b"A" * 70000,                          # oversized junk, no CRLF
+            b"GET /api/v1/query HTTP/1.1\r\nHost",  # truncated header
+        ]
+        for p in payloads:
+            s = socket.create_connection((host, port), timeout=5)
+            try:
+                s.sendall(p)
+                s.settimeout(2)
+                try:
+                    s.recv(4096)
+                except socket.timeout:
+                    pass
+            finally:
+                s.close()
+        # server still serves real requests afterwards
+        import urllib.request
+
+        r = urllib.request.urlopen(b.prom_url + "/api/v1/query?query=up", timeout=5)
+        assert r.status == 200
+    finally:
+        b.stop()
