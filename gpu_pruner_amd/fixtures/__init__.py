@@ -11,5 +11,7 @@ from .fake_prom import FakePrometheus
 from .fake_apiserver import FakeApiServer
 from .synth import build_synthetic_cluster
 from .fake_otlp import FakeOtlpCollector
+from .miniprom import MiniProm
 
-__all__ = ["FakePrometheus", "FakeApiServer", "build_synthetic_cluster", "FakeOtlpCollector"]
+__all__ = ["FakePrometheus", "FakeApiServer", "build_synthetic_cluster",
+           "FakeOtlpCollector", "MiniProm"]
