@@ -300,3 +300,80 @@ def test_busy_gpu_is_not_culled_idle_gpu_is():
     finally:
         backend.stop()
         sampler.stop()
+
+
+def test_window_semantics_with_real_gpu_signal(fake_api, pruner_bin):
+    """Three-tier pipeline on silicon: the exporter's real activity series
+    are scraped into MiniProm (which executes max_over_time like Prometheus)
+    and the daemon decides from them. A real probe burst inside the 1-minute
+    window protects the pod; once the burst ages out, the pod is culled."""
+    _require_gpu()
+    from gpu_pruner_amd.fixtures import MiniProm
+    from gpu_pruner_amd import probe
+
+    port = 19402
+    exporter = subprocess.Popen(
+        [str(REPO_ROOT / "bin" / "mi355-exporter"), "-p", str(port),
+         "-b", "127.0.0.1", "-i", "200"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+
+    def scrape_into(prom):
+        text = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=3).read().decode()
+        series = _parse_prom_text(text, "DCGM_FI_PROF_GR_ENGINE_ACTIVE")
+        gpu0 = [s for s in series if s[0].get("gpu") == "0"][0]
+        prom.ingest_activity("train-0", "ml", gpu0[1])
+        return gpu0[1]
+
+    dep = fake_api.add_deployment("train", "ml")
+    rs = fake_api.add_replicaset("train-rs", "ml", owner=dep)
+    fake_api.add_pod("train-0", "ml", owner_kind="ReplicaSet", owner_name="train-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+
+    with MiniProm() as prom:
+        try:
+            for _ in range(50):  # wait for the exporter
+                time.sleep(0.2)
+                try:
+                    urllib.request.urlopen(f"http://127.0.0.1:{port}/healthz",
+                                           timeout=2).read()
+                    break
+                except OSError:
+                    continue
+            # real burst: probe load, scraped into the store
+            with probe.busy_load(device=0, max_seconds=20.0):
+                time.sleep(1.5)
+                busy_val = scrape_into(prom)
+            assert busy_val > 0.0
+            # settle + scrape an idle sample
+            deadline = time.monotonic() + 15
+            idle_val = 1.0
+            while time.monotonic() < deadline:
+                time.sleep(1.0)
+                idle_val = scrape_into(prom)
+                if idle_val == 0.0:
+                    break
+            assert idle_val == 0.0
+
+            args = [pruner_bin, "--prometheus-url", prom.url, "--run-mode",
+                    "scale-down", "-t", "1", "--grace-period", "0"]
+            # burst still inside the 1-minute window → protected
+            r = subprocess.run(args, capture_output=True, text=True, timeout=30, env=env)
+            assert r.returncode == 0, r.stderr
+            assert fake_api.get("Deployment", "ml", "train")["spec"]["replicas"] == 1
+
+            # wait for the burst to age out of the window, keep scraping idle
+            for _ in range(70):
+                time.sleep(1.0)
+                if scrape_into(prom) != 0.0:
+                    continue
+            r = subprocess.run(args, capture_output=True, text=True, timeout=30, env=env)
+            assert r.returncode == 0, r.stderr
+            assert fake_api.get("Deployment", "ml", "train")["spec"]["replicas"] == 0
+        finally:
+            exporter.terminate()
+            exporter.wait(timeout=10)
