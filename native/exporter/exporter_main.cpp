@@ -39,6 +39,8 @@ OPTIONS:
   -p, --port <PORT>        listen port [default: 9400]
   -b, --bind <ADDR>        bind address [default: 0.0.0.0]
   -i, --interval <MS>      activity poll interval [default: 1000]
+      --idle-epsilon <R>   windowed activity ratios below R report as exactly
+                           0 (firmware housekeeping noise floor) [default: 0.005]
       --node-type <STR>    value for the node_type const label
       --hostname <STR>     override Hostname label (default: gethostname)
   -l, --log-format <FMT>   default | json | pretty
@@ -67,6 +69,7 @@ int main(int argc, char** argv) {
   uint16_t port = 9400;
   std::string bind_addr = "0.0.0.0";
   int interval_ms = 1000;
+  double idle_epsilon = 0.005;
   std::string node_type;
   std::string hostname;
   logx::Format fmt = logx::Format::Default;
@@ -83,6 +86,7 @@ int main(int argc, char** argv) {
     if (a == "-p" || a == "--port") port = static_cast<uint16_t>(std::stoi(next()));
     else if (a == "-b" || a == "--bind") bind_addr = next();
     else if (a == "-i" || a == "--interval") interval_ms = std::stoi(next());
+    else if (a == "--idle-epsilon") idle_epsilon = std::stod(next());
     else if (a == "--node-type") node_type = next();
     else if (a == "--hostname") hostname = next();
     else if (a == "-l" || a == "--log-format") {
@@ -109,7 +113,7 @@ int main(int argc, char** argv) {
     else if (::gethostname(buf, sizeof buf - 1) == 0) hostname = buf;
   }
 
-  exporter::Sampler sampler(interval_ms);
+  exporter::Sampler sampler(interval_ms, idle_epsilon);
   try {
     sampler.init();
   } catch (const std::exception& e) {

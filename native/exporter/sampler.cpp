@@ -56,7 +56,8 @@ uint64_t kfd_gpu_id_for_bdf(uint64_t bdfid) {
 
 }  // namespace
 
-Sampler::Sampler(int poll_interval_ms) : poll_interval_ms_(poll_interval_ms) {}
+Sampler::Sampler(int poll_interval_ms, double idle_epsilon)
+    : poll_interval_ms_(poll_interval_ms), idle_epsilon_(idle_epsilon) {}
 
 Sampler::~Sampler() {
   stop();
@@ -229,8 +230,11 @@ void Sampler::poll_device(uint32_t i) {
   st.prev_fw_ts = d.firmware_timestamp;
   st.have_prev_acc = d.metrics_table_ok;
 
-  d.gr_engine_active =
+  double window_ratio =
       st.wall_seconds > 0 ? std::min(st.busy_seconds / st.wall_seconds, 1.0) : 0.0;
+  // Floor sub-noise ratios to an exact 0 (see ctor comment): the idle
+  // predicate is `== 0`, and housekeeping blips are not workload.
+  d.gr_engine_active = window_ratio < idle_epsilon_ ? 0.0 : window_ratio;
 }
 
 std::vector<DeviceSample> Sampler::snapshot(bool reset_window) {

@@ -64,8 +64,14 @@ public:
 class Sampler {
 public:
   // poll_interval_ms: cadence of the background poll used to integrate the
-  // windowed activity ratio.
-  explicit Sampler(int poll_interval_ms = 1000);
+  // windowed activity ratio. idle_epsilon: windowed ratios below this are
+  // reported as exactly 0.0 — idle MI355X silicon emits sporadic
+  // firmware/driver housekeeping blips of ~0.02-0.03% activity
+  // (profiles/raw/winsem_debug.log) which would otherwise poison the
+  // culler's `== 0` predicate over long windows with false negatives (the
+  // same trap NVIDIA's DCGM PROF metrics have). 0.5% is far above the
+  // noise floor and far below any real workload.
+  explicit Sampler(int poll_interval_ms = 1000, double idle_epsilon = 0.005);
   ~Sampler();
 
   // Initialize rocm_smi and enumerate devices. Throws SamplerError when the
@@ -100,6 +106,7 @@ private:
   void poll_device(uint32_t i);
 
   int poll_interval_ms_;
+  double idle_epsilon_;
   uint32_t n_devices_ = 0;
   bool initialized_ = false;
   std::mutex mu_;

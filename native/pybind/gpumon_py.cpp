@@ -76,7 +76,8 @@ PYBIND11_MODULE(_gpumon, m) {
   py::register_exception<SamplerError>(m, "SamplerError");
 
   py::class_<Sampler>(m, "Sampler")
-      .def(py::init<int>(), py::arg("poll_interval_ms") = 1000)
+      .def(py::init<int, double>(), py::arg("poll_interval_ms") = 1000,
+           py::arg("idle_epsilon") = 0.005)
       .def("init", &Sampler::init,
            "Initialize rocm_smi and enumerate devices (raises SamplerError "
            "without an AMD GPU)")
