@@ -55,7 +55,8 @@ def _merge_patch(target, patch):
 
 class FakeApiServer:
     def __init__(self, host: str = "127.0.0.1", port: int = 0, token: str | None = None,
-                 latency_s: float = 0.0):
+                 latency_s: float = 0.0, certfile: str | None = None,
+                 keyfile: str | None = None, client_ca: str | None = None):
         self._lock = threading.Lock()
         # objects[(kind, namespace, name)] = dict
         self.objects: dict[tuple[str, str, str], dict] = {}
@@ -191,6 +192,16 @@ class FakeApiServer:
 
         self._server = ThreadingHTTPServer((host, port), Handler)
         self._server.daemon_threads = True
+        self._tls = certfile is not None
+        if certfile is not None:
+            import ssl
+
+            ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            ctx.load_cert_chain(certfile, keyfile)
+            if client_ca is not None:  # mTLS: require a client certificate
+                ctx.load_verify_locations(client_ca)
+                ctx.verify_mode = ssl.CERT_REQUIRED
+            self._server.socket = ctx.wrap_socket(self._server.socket, server_side=True)
         self._thread = threading.Thread(
             target=lambda: self._server.serve_forever(poll_interval=0.05), daemon=True)
 
@@ -206,7 +217,8 @@ class FakeApiServer:
     @property
     def url(self) -> str:
         host, port = self._server.server_address[:2]
-        return f"http://{host}:{port}"
+        scheme = "https" if self._tls else "http"
+        return f"{scheme}://{host}:{port}"
 
     def __enter__(self):
         return self.start()

@@ -312,6 +312,17 @@ Client::Client(Url base, ClientOptions opts) : base_(std::move(base)), opts_(std
         }
         break;
     }
+    // mTLS client authentication (kube client-certificate auth)
+    if (!opts_.client_cert_file.empty()) {
+      if (SSL_CTX_use_certificate_chain_file(ctx, opts_.client_cert_file.c_str()) != 1 ||
+          SSL_CTX_use_PrivateKey_file(ctx, opts_.client_key_file.c_str(),
+                                      SSL_FILETYPE_PEM) != 1 ||
+          SSL_CTX_check_private_key(ctx) != 1) {
+        SSL_CTX_free(ctx);
+        throw Error("failed to load client certificate/key (" + opts_.client_cert_file +
+                    ", " + opts_.client_key_file + "): " + ssl_err_string());
+      }
+    }
     ssl_ctx_ = ctx;
   }
 }

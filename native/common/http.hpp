@@ -51,9 +51,11 @@ struct Response {
 
 struct ClientOptions {
   TlsVerify tls = TlsVerify::Verify;
-  std::string ca_file;         // PEM bundle for TlsVerify::CustomCa
+  std::string ca_file;          // PEM bundle for TlsVerify::CustomCa
+  std::string client_cert_file; // mTLS client certificate (PEM), optional
+  std::string client_key_file;  // mTLS client private key (PEM)
   int connect_timeout_ms = 5000;
-  int io_timeout_ms = 30000;   // per-request read/write deadline
+  int io_timeout_ms = 30000;    // per-request read/write deadline
   int max_pool_per_origin = 256;
 };
 

@@ -31,6 +31,8 @@ KubeConfig KubeConfig::resolve() {
     if (const char* t = env("GPU_PRUNER_K8S_TOKEN")) cfg.token = t;
     if (const char* tf = env("GPU_PRUNER_K8S_TOKEN_FILE")) cfg.token_file = tf;
     if (const char* ca = env("GPU_PRUNER_K8S_CA")) cfg.ca_file = ca;
+    if (const char* cc = env("GPU_PRUNER_K8S_CLIENT_CERT")) cfg.client_cert_file = cc;
+    if (const char* ck = env("GPU_PRUNER_K8S_CLIENT_KEY")) cfg.client_key_file = ck;
     if (const char* skip = env("GPU_PRUNER_K8S_SKIP_TLS"))
       cfg.skip_tls = std::string(skip) != "0" && strutil::lower(skip) != "false";
     if (const char* ns = env("GPU_PRUNER_K8S_NAMESPACE")) cfg.default_namespace = ns;
@@ -66,6 +68,10 @@ KubeClient::KubeClient(KubeConfig cfg) : cfg_(std::move(cfg)) {
   } else if (cfg_.ca_file) {
     opts.tls = http::TlsVerify::CustomCa;
     opts.ca_file = *cfg_.ca_file;
+  }
+  if (cfg_.client_cert_file && cfg_.client_key_file) {
+    opts.client_cert_file = *cfg_.client_cert_file;
+    opts.client_key_file = *cfg_.client_key_file;
   }
   http_ = std::make_unique<http::Client>(*url, opts);
 }

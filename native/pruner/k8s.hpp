@@ -29,6 +29,9 @@ struct KubeConfig {
   std::optional<std::string> token;
   std::optional<std::string> token_file;  // re-read per client build (token rotation)
   std::optional<std::string> ca_file;
+  // mTLS client-certificate auth (kube client-cert users); both required
+  std::optional<std::string> client_cert_file;
+  std::optional<std::string> client_key_file;
   bool skip_tls = false;
   std::string default_namespace = "default";
 

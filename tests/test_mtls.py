@@ -1,0 +1,92 @@
+"""mTLS client-certificate auth against the apiserver.
+
+Kubernetes clusters commonly authenticate out-of-cluster clients with
+client certificates (kubeconfig `client-certificate`/`client-key`); the
+native HTTP client supports this via GPU_PRUNER_K8S_CLIENT_CERT/KEY
+(native/common/http.cpp SSL_CTX_use_certificate_chain_file path).
+"""
+
+import os
+import subprocess
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def pki(tmp_path_factory):
+    """CA + server cert (SAN 127.0.0.1) + client cert signed by the CA."""
+    d = tmp_path_factory.mktemp("mtls")
+
+    def run(*args):
+        subprocess.run(["openssl", *args], check=True, capture_output=True)
+
+    ca_key, ca_crt = d / "ca.key", d / "ca.crt"
+    run("req", "-x509", "-newkey", "rsa:2048", "-nodes", "-keyout", str(ca_key),
+        "-out", str(ca_crt), "-days", "2", "-subj", "/CN=test-ca")
+
+    def issue(name, cn, san=None):
+        key, csr, crt = d / f"{name}.key", d / f"{name}.csr", d / f"{name}.crt"
+        run("req", "-newkey", "rsa:2048", "-nodes", "-keyout", str(key),
+            "-out", str(csr), "-subj", f"/CN={cn}")
+        ext = []
+        if san:
+            extfile = d / f"{name}.ext"
+            extfile.write_text(f"subjectAltName={san}\n")
+            ext = ["-extfile", str(extfile)]
+        run("x509", "-req", "-in", str(csr), "-CA", str(ca_crt), "-CAkey", str(ca_key),
+            "-CAcreateserial", "-out", str(crt), "-days", "2", *ext)
+        return key, crt
+
+    s_key, s_crt = issue("server", "127.0.0.1", san="IP:127.0.0.1")
+    c_key, c_crt = issue("client", "gpu-pruner-user")
+    return {"ca": str(ca_crt), "server_key": str(s_key), "server_crt": str(s_crt),
+            "client_key": str(c_key), "client_crt": str(c_crt)}
+
+
+@pytest.fixture
+def mtls_api(pki):
+    from gpu_pruner_amd.fixtures import FakeApiServer
+
+    with FakeApiServer(certfile=pki["server_crt"], keyfile=pki["server_key"],
+                       client_ca=pki["ca"]) as a:
+        yield a
+
+
+def run_pruner(pruner_bin, api_url, prom_url, pki, with_client_cert=True):
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = api_url
+    env["GPU_PRUNER_K8S_CA"] = pki["ca"]
+    env.pop("GPU_PRUNER_K8S_SKIP_TLS", None)
+    if with_client_cert:
+        env["GPU_PRUNER_K8S_CLIENT_CERT"] = pki["client_crt"]
+        env["GPU_PRUNER_K8S_CLIENT_KEY"] = pki["client_key"]
+    else:
+        env.pop("GPU_PRUNER_K8S_CLIENT_CERT", None)
+        env.pop("GPU_PRUNER_K8S_CLIENT_KEY", None)
+    env["PROMETHEUS_TOKEN"] = "t"
+    return subprocess.run(
+        [pruner_bin, "--prometheus-url", prom_url, "--run-mode", "scale-down"],
+        capture_output=True, text=True, timeout=30, env=env)
+
+
+def test_mtls_apiserver_full_cull(pruner_bin, mtls_api, fake_prom, pki):
+    dep = mtls_api.add_deployment("d", "ml")
+    rs = mtls_api.add_replicaset("d-rs", "ml", owner=dep)
+    mtls_api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p0", "ml")
+    r = run_pruner(pruner_bin, mtls_api.url, fake_prom.url, pki)
+    assert r.returncode == 0, r.stderr
+    assert mtls_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0
+
+
+def test_mtls_without_client_cert_rejected(pruner_bin, mtls_api, fake_prom, pki):
+    """The server requires a client cert: handshake fails, pods are skipped."""
+    dep = mtls_api.add_deployment("d", "ml")
+    rs = mtls_api.add_replicaset("d-rs", "ml", owner=dep)
+    mtls_api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p0", "ml")
+    r = run_pruner(pruner_bin, mtls_api.url, fake_prom.url, pki, with_client_cert=False)
+    assert mtls_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 1
+    assert "TLS" in r.stderr or "error" in r.stderr.lower()
