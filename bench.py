@@ -78,8 +78,15 @@ def main():
         # Control-plane workload: no tensor collectives exist in this
         # framework (SURVEY.md §5.8 — the reference has no data plane), so
         # the barrier/reduce backend is gloo; CUDA work is still synchronized
-        # per-rank below.
-        dist.init_process_group(backend="gloo")
+        # per-rank below. Gloo prints "[Gloo] Rank ..." on fd 1 — divert it
+        # to stderr so stdout stays exactly one JSON line (driver contract).
+        saved_stdout = os.dup(1)
+        try:
+            os.dup2(2, 1)
+            dist.init_process_group(backend="gloo")
+        finally:
+            os.dup2(saved_stdout, 1)
+            os.close(saved_stdout)
 
     have_cuda = torch.cuda.is_available()
     if have_cuda:
