@@ -25,7 +25,9 @@ def test_bench_two_ranks_gloo():
          "--gpus", "2", "--steps", "3", "--warmup", "1", "--pods", "100"],
         capture_output=True, text=True, timeout=300, cwd=str(REPO_ROOT), env=env)
     assert r.returncode == 0, r.stderr[-3000:]
-    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    all_lines = r.stdout.strip().splitlines()
+    assert len(all_lines) == 1, f"stdout must be exactly one line, got: {r.stdout!r}"
+    lines = [l for l in all_lines if l.startswith("{")]
     assert len(lines) == 1, f"exactly one JSON line expected, got: {r.stdout!r}"
     result = json.loads(lines[0])
     assert result["n_gpus"] == 2
