@@ -169,6 +169,9 @@ def main():
     t_end = time.perf_counter()
 
     elapsed = t_end - t_start
+    log(f"[bench] rank {rank}: elapsed {elapsed:.3f}s "
+        f"p50 {statistics.median(step_times)*1000:.1f}ms "
+        f"max {max(step_times)*1000:.1f}ms")
     # MAX elapsed over ranks (driver contract)
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64)

@@ -139,13 +139,17 @@ void Sampler::stop() {
 }
 
 void Sampler::poll_once() {
+  std::lock_guard<std::mutex> poll_lock(poll_mu_);
   for (uint32_t i = 0; i < n_devices_; i++) poll_device(i);
 }
 
 void Sampler::poll_device(uint32_t i) {
-  std::lock_guard<std::mutex> lock(mu_);
-  DevState& st = devices_[i];
-  DeviceSample& d = st.last;
+  // rsmi/SMU reads happen on a local copy, without holding the state lock
+  DeviceSample d;
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    d = devices_[i].last;
+  }
   double now = monotonic_s();
 
   uint32_t busy = 0;
@@ -197,7 +201,9 @@ void Sampler::poll_device(uint32_t i) {
       d.busy_percent = gm.average_gfx_activity;
   }
 
-  // ---- integrate the scrape-window activity ratio ----
+  // ---- integrate the scrape-window activity ratio (brief state lock) ----
+  std::lock_guard<std::mutex> lock(mu_);
+  DevState& st = devices_[i];
   if (st.prev_poll_monotonic > 0.0) {
     double dt = now - st.prev_poll_monotonic;
     if (dt > 0) {
@@ -235,6 +241,7 @@ void Sampler::poll_device(uint32_t i) {
   // Floor sub-noise ratios to an exact 0 (see ctor comment): the idle
   // predicate is `== 0`, and housekeeping blips are not workload.
   d.gr_engine_active = window_ratio < idle_epsilon_ ? 0.0 : window_ratio;
+  st.last = d;
 }
 
 std::vector<DeviceSample> Sampler::snapshot(bool reset_window) {

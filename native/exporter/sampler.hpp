@@ -109,6 +109,11 @@ private:
   double idle_epsilon_;
   uint32_t n_devices_ = 0;
   bool initialized_ = false;
+  // poll_mu_ serializes pollers; mu_ guards only the stored state so
+  // snapshot() never waits behind a slow firmware (SMU) read — concurrent
+  // rsmi gpu_metrics reads from several processes can stall ~80 ms
+  // (observed under the 4-rank bench) and must not block consumers.
+  std::mutex poll_mu_;
   std::mutex mu_;
   std::vector<DevState> devices_;
   std::thread poller_;
