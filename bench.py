@@ -57,9 +57,9 @@ def main():
                     help="total synthetic pods across all ranks")
     ap.add_argument("--latency-us", type=int, default=0,
                     help="injected apiserver latency per request (RTT emulation)")
-    ap.add_argument("--concurrency", type=int,
-                    default=int(os.environ.get("BENCH_CONCURRENCY", "32")),
-                    help="engine --max-concurrency")
+    ap.add_argument("--concurrency", type=int, default=None,
+                    help="engine --max-concurrency (default: 32 split across "
+                         "co-located ranks)")
     ap.add_argument("--otlp", action="store_true",
                     help="export spans/metrics to an in-process OTLP collector "
                          "during the timed region (BASELINE config 5)")
@@ -72,6 +72,14 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world_size > 1
+
+    # Co-located ranks share one node's CPU allowance: split the engine
+    # fan-out and thread pool across them or CFS throttling synchronizes
+    # ~80 ms stalls into every rank (measured; see threadpool.hpp).
+    if args.concurrency is None:
+        args.concurrency = int(os.environ.get("BENCH_CONCURRENCY",
+                                              str(max(8, 32 // world_size))))
+    os.environ.setdefault("GPU_PRUNER_POOL_SIZE", str(max(32, 256 // world_size)))
 
     n_gpus = world_size if distributed else args.gpus
     if distributed:

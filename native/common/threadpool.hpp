@@ -9,6 +9,7 @@
 #pragma once
 
 #include <atomic>
+#include <cstdio>
 #include <cstdlib>
 #include <condition_variable>
 #include <deque>
@@ -92,16 +93,58 @@ private:
     }
   }
 
+  // Effective CPU allowance: the cgroup quota when set (containers often
+  // cap far below the node's core count — a 256-core box with a 16-CPU
+  // quota CFS-throttles oversized thread herds into synchronized ~80 ms
+  // stalls), else hardware_concurrency.
+  static size_t allowed_cpus() {
+    auto read_quota = [](const char* path, const char* fmt) -> double {
+      if (FILE* f = std::fopen(path, "r")) {
+        long long quota = -1, period = -1;
+        char buf[64] = {0};
+        if (std::fgets(buf, sizeof buf, f)) {
+          if (buf[0] == 'm') {  // "max <period>" = no quota
+            std::fclose(f);
+            return 0;
+          }
+          std::sscanf(buf, fmt, &quota, &period);
+        }
+        std::fclose(f);
+        if (quota > 0 && period > 0) return static_cast<double>(quota) / period;
+      }
+      return 0;
+    };
+    double cpus = read_quota("/sys/fs/cgroup/cpu.max", "%lld %lld");  // v2
+    if (cpus <= 0) {  // v1 pair
+      long long q = -1, p = -1;
+      if (FILE* f = std::fopen("/sys/fs/cgroup/cpu/cpu.cfs_quota_us", "r")) {
+        std::fscanf(f, "%lld", &q);
+        std::fclose(f);
+      }
+      if (FILE* f = std::fopen("/sys/fs/cgroup/cpu/cpu.cfs_period_us", "r")) {
+        std::fscanf(f, "%lld", &p);
+        std::fclose(f);
+      }
+      if (q > 0 && p > 0) cpus = static_cast<double>(q) / p;
+    }
+    if (cpus <= 0) cpus = static_cast<double>(std::thread::hardware_concurrency());
+    return cpus < 1 ? 1 : static_cast<size_t>(cpus);
+  }
+
   static size_t default_size() {
-    // The fan-out is I/O-bound (apiserver round-trips), so worker count
-    // follows the maximum useful request concurrency, not the core count —
-    // idle workers just sleep on the condition variable. Overridable for
-    // constrained deployments.
+    // The fan-out is I/O-bound (apiserver round-trips), so worker count can
+    // exceed the CPU allowance — blocked threads cost no quota — but herds
+    // far beyond it burn quota on wakeups and trip CFS throttling. 4× the
+    // allowance balances both; override with GPU_PRUNER_POOL_SIZE.
     if (const char* env = std::getenv("GPU_PRUNER_POOL_SIZE"); env && *env) {
       long v = std::strtol(env, nullptr, 10);
       if (v > 0 && v <= 4096) return static_cast<size_t>(v);
     }
-    return 256;
+    size_t cpus = allowed_cpus();
+    size_t size = cpus * 4;
+    if (size < 32) size = 32;
+    if (size > 256) size = 256;
+    return size;
   }
 
   void worker_loop() {
