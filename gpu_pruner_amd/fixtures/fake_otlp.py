@@ -16,6 +16,8 @@ class FakeOtlpCollector:
         self._lock = threading.Lock()
         self.traces: list[dict] = []
         self.metrics: list[dict] = []
+        self.traces_pb: list[bytes] = []   # raw binary-protobuf payloads
+        self.metrics_pb: list[bytes] = []
 
         fixture = self
 
@@ -28,15 +30,26 @@ class FakeOtlpCollector:
 
             def do_POST(self):
                 length = int(self.headers.get("Content-Length", "0"))
-                try:
-                    payload = json.loads(self.rfile.read(length) or b"{}")
-                except json.JSONDecodeError:
-                    payload = {}
-                with fixture._lock:
-                    if self.path.endswith("/v1/traces"):
-                        fixture.traces.append(payload)
-                    elif self.path.endswith("/v1/metrics"):
-                        fixture.metrics.append(payload)
+                raw = self.rfile.read(length) or b""
+                ctype = self.headers.get("Content-Type", "")
+                if "protobuf" in ctype:
+                    with fixture._lock:
+                        if self.path.endswith("/v1/traces"):
+                            fixture.traces_pb.append(raw)
+                        elif self.path.endswith("/v1/metrics"):
+                            fixture.metrics_pb.append(raw)
+                    payload = None
+                else:
+                    try:
+                        payload = json.loads(raw or b"{}")
+                    except json.JSONDecodeError:
+                        payload = {}
+                if payload is not None:
+                    with fixture._lock:
+                        if self.path.endswith("/v1/traces"):
+                            fixture.traces.append(payload)
+                        elif self.path.endswith("/v1/metrics"):
+                            fixture.metrics.append(payload)
                 body = b"{}"
                 self.send_response(200)
                 self.send_header("Content-Type", "application/json")

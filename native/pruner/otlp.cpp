@@ -9,6 +9,7 @@
 #include <vector>
 
 #include "../common/http.hpp"
+#include "../common/pb.hpp"
 #include "../common/json.hpp"
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
@@ -40,6 +41,7 @@ struct State {
   std::condition_variable cv;
   bool stop = false;
   int interval_ms = 5000;
+  bool use_protobuf = false;  // OTEL_EXPORTER_OTLP_PROTOCOL=http/protobuf
   std::unique_ptr<http::Client> client;  // persistent export connection pool
 };
 
@@ -59,6 +61,102 @@ std::string rand_hex(size_t bytes) {
   return id.substr(0, bytes * 2);
 }
 
+std::string hex_to_bytes(const std::string& hex) {
+  std::string out;
+  out.reserve(hex.size() / 2);
+  for (size_t i = 0; i + 1 < hex.size(); i += 2) {
+    auto nib = [](char c) -> int {
+      if (c >= '0' && c <= '9') return c - '0';
+      if (c >= 'a' && c <= 'f') return c - 'a' + 10;
+      if (c >= 'A' && c <= 'F') return c - 'A' + 10;
+      return 0;
+    };
+    out += static_cast<char>((nib(hex[i]) << 4) | nib(hex[i + 1]));
+  }
+  return out;
+}
+
+// ---- binary-protobuf encoders (opentelemetry-proto field numbers) ----------
+
+std::string pb_resource(const std::string& service_name) {
+  std::string any;  // AnyValue{string_value=1}
+  pb::put_bytes(any, 1, service_name);
+  std::string kv;  // KeyValue{key=1, value=2}
+  pb::put_bytes(kv, 1, "service.name");
+  pb::put_bytes(kv, 2, any);
+  std::string res;  // Resource{attributes=1}
+  pb::put_bytes(res, 1, kv);
+  return res;
+}
+
+std::string pb_scope() {
+  std::string scope;  // InstrumentationScope{name=1}
+  pb::put_bytes(scope, 1, "gpu_pruner::main");
+  return scope;
+}
+
+// ExportTraceServiceRequest{resource_spans=1{resource=1, scope_spans=2{scope=1, spans=2}}}
+std::string encode_spans_pb(const State& s, const std::vector<FinishedSpan>& batch) {
+  std::string spans;
+  for (const auto& fs : batch) {
+    std::string sp;  // Span
+    pb::put_bytes(sp, 1, hex_to_bytes(s.trace_id));  // trace_id (16 bytes)
+    pb::put_bytes(sp, 2, hex_to_bytes(rand_hex(8))); // span_id (8 bytes)
+    pb::put_bytes(sp, 5, fs.name);                   // name
+    pb::put_varint(sp, 6, 1);                        // kind = INTERNAL
+    pb::put_fixed64(sp, 7, fs.start_ns);
+    pb::put_fixed64(sp, 8, fs.end_ns);
+    pb::put_bytes(spans, 2, sp);                     // ScopeSpans.spans
+  }
+  std::string scope_spans;
+  pb::put_bytes(scope_spans, 1, pb_scope());
+  scope_spans += spans;
+  std::string rs;  // ResourceSpans
+  pb::put_bytes(rs, 1, pb_resource(s.service_name));
+  pb::put_bytes(rs, 2, scope_spans);
+  std::string req;
+  pb::put_bytes(req, 1, rs);
+  return req;
+}
+
+// ExportMetricsServiceRequest{resource_metrics=1{resource=1, scope_metrics=2{scope=1, metrics=2}}}
+std::string encode_metrics_pb(const State& s,
+                              const std::map<std::string, int64_t>& counters,
+                              uint64_t t_ns) {
+  std::string metrics;
+  for (const auto& [name, value] : counters) {
+    bool monotonic = strutil::starts_with(name, "monotonic_counter.");
+    std::string short_name = name.substr(name.find('.') + 1);
+    std::string dp;  // NumberDataPoint{start=2, time=3, as_int=6 (sfixed64)}
+    if (monotonic) pb::put_fixed64(dp, 2, t_ns);
+    pb::put_fixed64(dp, 3, t_ns);
+    pb::put_fixed64(dp, 6, static_cast<uint64_t>(value));
+    std::string m;  // Metric{name=1, gauge=5 | sum=7}
+    pb::put_bytes(m, 1, short_name);
+    if (monotonic) {
+      std::string sum;  // Sum{data_points=1, temporality=2, is_monotonic=3}
+      pb::put_bytes(sum, 1, dp);
+      pb::put_varint(sum, 2, 2);  // CUMULATIVE
+      pb::put_varint(sum, 3, 1);
+      pb::put_bytes(m, 7, sum);
+    } else {
+      std::string gauge;  // Gauge{data_points=1}
+      pb::put_bytes(gauge, 1, dp);
+      pb::put_bytes(m, 5, gauge);
+    }
+    pb::put_bytes(metrics, 2, m);  // ScopeMetrics.metrics
+  }
+  std::string scope_metrics;
+  pb::put_bytes(scope_metrics, 1, pb_scope());
+  scope_metrics += metrics;
+  std::string rm;
+  pb::put_bytes(rm, 1, pb_resource(s.service_name));
+  pb::put_bytes(rm, 2, scope_metrics);
+  std::string req;
+  pb::put_bytes(req, 1, rm);
+  return req;
+}
+
 jsn::Value resource_json(const std::string& service_name) {
   jsn::Value attr = jsn::Value::object();
   attr["key"] = "service.name";
@@ -69,7 +167,7 @@ jsn::Value resource_json(const std::string& service_name) {
   return res;
 }
 
-void post_json(const std::string& url, const jsn::Value& body) {
+void post_payload(const std::string& url, std::string body, const char* content_type) {
   State& s = state();
   auto parsed = http::Url::parse(url);
   if (!parsed) return;
@@ -82,11 +180,15 @@ void post_json(const std::string& url, const jsn::Value& body) {
   http::Request req;
   req.method = "POST";
   req.path = parsed->path;
-  req.body = body.dump();
-  req.headers.emplace_back("Content-Type", "application/json");
+  req.body = std::move(body);
+  req.headers.emplace_back("Content-Type", content_type);
   http::Response resp = s.client->request(req);
   if (resp.status >= 200 && resp.status < 300)
     s.delivered.fetch_add(1, std::memory_order_relaxed);
+}
+
+void post_json(const std::string& url, const jsn::Value& body) {
+  post_payload(url, body.dump(), "application/json");
 }
 
 void export_once() {
@@ -101,6 +203,13 @@ void export_once() {
     size_t n = std::min(kMaxSpansPerPost, all.size() - base);
     std::vector<FinishedSpan> batch(all.begin() + static_cast<long>(base),
                                     all.begin() + static_cast<long>(base + n));
+    if (s.use_protobuf) {
+      try {
+        post_payload(s.endpoint + "/v1/traces", encode_spans_pb(s, batch),
+                     "application/x-protobuf");
+      } catch (const std::exception&) { /* collector away; drop batch */ }
+      continue;
+    }
     jsn::Value spans = jsn::Value::array();
     for (const auto& fs : batch) {
       jsn::Value sp = jsn::Value::object();
@@ -128,7 +237,13 @@ void export_once() {
 
   // ---- metrics: the counter registry (monotonic counters + gauges) ----
   auto counters = logx::counters_snapshot();
-  if (!counters.empty()) {
+  if (!counters.empty() && s.use_protobuf) {
+    try {
+      post_payload(s.endpoint + "/v1/metrics",
+                   encode_metrics_pb(s, counters, now_unix_ns()),
+                   "application/x-protobuf");
+    } catch (const std::exception&) { /* collector away */ }
+  } else if (!counters.empty()) {
     uint64_t t = now_unix_ns();
     jsn::Value metrics = jsn::Value::array();
     for (const auto& [name, value] : counters) {
@@ -181,6 +296,8 @@ void init(const std::string& service_name) {
   s.trace_id = rand_hex(16);
   if (const char* iv = std::getenv("OTEL_METRIC_EXPORT_INTERVAL"); iv && *iv)
     s.interval_ms = std::atoi(iv);
+  if (const char* proto = std::getenv("OTEL_EXPORTER_OTLP_PROTOCOL"); proto && *proto)
+    s.use_protobuf = std::string(proto) == "http/protobuf";
   s.enabled.store(true);
   s.running.store(true);
   s.exporter = std::thread([&s] {
