@@ -3,7 +3,11 @@
 #include <fstream>
 
 #include "../common/log.hpp"
+#include "../common/miniyaml.hpp"
 #include "../common/strutil.hpp"
+
+#include <stdlib.h>
+#include <unistd.h>
 
 namespace pruner {
 
@@ -19,6 +23,89 @@ std::optional<std::string> read_file(const std::string& path) {
 const char* env(const char* name) {
   const char* v = std::getenv(name);
   return v && *v ? v : nullptr;
+}
+
+// Write decoded -data material to a private temp file; returns the path.
+std::string write_temp_pem(const std::string& data, const char* tag) {
+  std::string tmpl = std::string("/tmp/gpu-pruner-") + tag + "-XXXXXX";
+  std::vector<char> buf(tmpl.begin(), tmpl.end());
+  buf.push_back('\0');
+  int fd = ::mkstemp(buf.data());
+  if (fd < 0) throw std::runtime_error("mkstemp failed for kubeconfig material");
+  ssize_t n = ::write(fd, data.data(), data.size());
+  ::close(fd);
+  if (n != static_cast<ssize_t>(data.size()))
+    throw std::runtime_error("short write of kubeconfig material");
+  return std::string(buf.data());
+}
+
+// Minimal kubeconfig loader (current-context → cluster + user). The
+// reference gets this via kube-rs Config::infer; here a YAML-subset parser
+// (common/miniyaml.hpp) covers kubectl-generated files: server, CA
+// (file or -data), insecure-skip-tls-verify, bearer token, client cert/key
+// (file or -data).
+std::optional<KubeConfig> load_kubeconfig(const std::string& path) {
+  std::ifstream f(path);
+  if (!f) return std::nullopt;
+  std::string src((std::istreambuf_iterator<char>(f)), std::istreambuf_iterator<char>());
+  jsn::Value doc;
+  try {
+    doc = miniyaml::parse(src);
+  } catch (const std::exception& e) {
+    LOGW("pruner::k8s", "failed to parse kubeconfig " + path + ": " + e.what());
+    return std::nullopt;
+  }
+
+  std::string ctx_name = doc.get("current-context").as_string();
+  auto find_named = [&](const char* list_key, const std::string& name) -> jsn::Value {
+    const jsn::Value& list = doc.get(list_key);
+    if (list.is_array())
+      for (const auto& item : list.arr())
+        if (item.get("name").as_string() == name) return item;
+    return jsn::Value();
+  };
+
+  jsn::Value ctx = find_named("contexts", ctx_name).get("context");
+  if (!ctx.is_object()) {
+    LOGW("pruner::k8s", "kubeconfig " + path + ": current-context not found");
+    return std::nullopt;
+  }
+  jsn::Value cluster = find_named("clusters", ctx.get("cluster").as_string()).get("cluster");
+  jsn::Value user = find_named("users", ctx.get("user").as_string()).get("user");
+  if (!cluster.is_object()) return std::nullopt;
+
+  KubeConfig cfg;
+  cfg.url = cluster.get("server").as_string();
+  if (cfg.url.empty()) return std::nullopt;
+  if (cluster.get("insecure-skip-tls-verify").as_bool(false)) cfg.skip_tls = true;
+  if (cluster.get("certificate-authority").is_string())
+    cfg.ca_file = cluster.get("certificate-authority").as_string();
+  else if (cluster.get("certificate-authority-data").is_string()) {
+    std::string pem;
+    if (miniyaml::base64_decode(cluster.get("certificate-authority-data").as_string(), &pem))
+      cfg.ca_file = write_temp_pem(pem, "ca");
+  }
+  if (user.is_object()) {
+    if (user.get("token").is_string()) cfg.token = user.get("token").as_string();
+    if (user.get("tokenFile").is_string()) cfg.token_file = user.get("tokenFile").as_string();
+    if (user.get("client-certificate").is_string())
+      cfg.client_cert_file = user.get("client-certificate").as_string();
+    else if (user.get("client-certificate-data").is_string()) {
+      std::string pem;
+      if (miniyaml::base64_decode(user.get("client-certificate-data").as_string(), &pem))
+        cfg.client_cert_file = write_temp_pem(pem, "cert");
+    }
+    if (user.get("client-key").is_string())
+      cfg.client_key_file = user.get("client-key").as_string();
+    else if (user.get("client-key-data").is_string()) {
+      std::string pem;
+      if (miniyaml::base64_decode(user.get("client-key-data").as_string(), &pem))
+        cfg.client_key_file = write_temp_pem(pem, "key");
+    }
+  }
+  if (ctx.get("namespace").is_string())
+    cfg.default_namespace = ctx.get("namespace").as_string();
+  return cfg;
 }
 
 }  // namespace
@@ -39,6 +126,14 @@ KubeConfig KubeConfig::resolve() {
     return cfg;
   }
 
+  // kubeconfig (like kube-rs Config::infer): $KUBECONFIG, then
+  // ~/.kube/config, before falling back to in-cluster.
+  if (const char* kc = env("KUBECONFIG")) {
+    if (auto cfg = load_kubeconfig(kc)) return *cfg;
+  } else if (const char* home = env("HOME")) {
+    if (auto cfg = load_kubeconfig(std::string(home) + "/.kube/config")) return *cfg;
+  }
+
   std::string sa_dir = "/var/run/secrets/kubernetes.io/serviceaccount";
   if (const char* dir = env("GPU_PRUNER_SA_DIR")) sa_dir = dir;
 
@@ -55,7 +150,8 @@ KubeConfig KubeConfig::resolve() {
   }
 
   throw std::runtime_error(
-      "no Kubernetes config: set GPU_PRUNER_K8S_URL or run in-cluster "
+      "no Kubernetes config: set GPU_PRUNER_K8S_URL, provide a kubeconfig "
+      "($KUBECONFIG / ~/.kube/config), or run in-cluster "
       "(KUBERNETES_SERVICE_HOST)");
 }
 

@@ -6,6 +6,7 @@
 
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
+#include "k8s.hpp"
 
 namespace pruner {
 
@@ -42,6 +43,23 @@ std::string get_prometheus_token() {
   if (const char* t = std::getenv("GPU_PRUNER_K8S_TOKEN"); t && *t) {
     LOGI(TARGET, "Using Kubernetes token for Prometheus");
     return t;
+  }
+  // token from the inferred kube config (kubeconfig user token / token file
+  // — reference lib.rs:210-222 reads the same via kube::Config::infer)
+  try {
+    KubeConfig kc = KubeConfig::resolve();
+    if (kc.token_file) {
+      if (auto t = read_file(*kc.token_file)) {
+        LOGI(TARGET, "Inferred Prometheus token from kube config token file");
+        return strutil::trim(*t);
+      }
+    }
+    if (kc.token) {
+      LOGI(TARGET, "Found K8s token");
+      return *kc.token;
+    }
+  } catch (const std::exception&) {
+    // no kube config at all — fall through to oc
   }
 
   // Last resort: the logged-in OpenShift user's token.

@@ -23,10 +23,13 @@ def sa_dir(tmp_path):
 def clean_env(monkeypatch):
     for var in ("GPU_PRUNER_K8S_URL", "GPU_PRUNER_K8S_TOKEN",
                 "GPU_PRUNER_K8S_TOKEN_FILE", "GPU_PRUNER_K8S_CA",
+                "GPU_PRUNER_K8S_CLIENT_CERT", "GPU_PRUNER_K8S_CLIENT_KEY",
                 "GPU_PRUNER_K8S_SKIP_TLS", "KUBERNETES_SERVICE_HOST",
                 "KUBERNETES_SERVICE_PORT", "GPU_PRUNER_SA_DIR",
-                "PROMETHEUS_TOKEN"):
+                "PROMETHEUS_TOKEN", "KUBECONFIG"):
         monkeypatch.delenv(var, raising=False)
+    # a stray ~/.kube/config must not leak into hermetic tests
+    monkeypatch.setenv("HOME", "/nonexistent-home")
     return monkeypatch
 
 
@@ -93,3 +96,96 @@ def test_prom_token_oc_whoami_last(core, clean_env, tmp_path, monkeypatch):
     oc.chmod(0o755)
     monkeypatch.setenv("PATH", f"{tmp_path}:/usr/bin:/bin")
     assert core.get_prometheus_token() == "oc-token-456"
+
+
+# ---- kubeconfig ($KUBECONFIG / ~/.kube/config) ------------------------------
+
+
+KUBECONFIG_TMPL = """\
+apiVersion: v1
+kind: Config
+current-context: prod
+clusters:
+- name: prod-cluster
+  cluster:
+    server: https://api.prod.example:6443
+    certificate-authority-data: {ca_b64}
+- name: other
+  cluster:
+    server: https://other:6443
+contexts:
+- name: prod
+  context:
+    cluster: prod-cluster
+    user: admin
+    namespace: ml-team
+- name: stale
+  context:
+    cluster: other
+    user: admin
+users:
+- name: admin
+  user:
+    token: kubeconfig-token-123
+"""
+
+
+def test_kubeconfig_token_auth(core, clean_env, tmp_path):
+    import base64
+
+    ca_pem = "-----BEGIN CERTIFICATE-----\nabc\n-----END CERTIFICATE-----\n"
+    kc = tmp_path / "config"
+    kc.write_text(KUBECONFIG_TMPL.format(
+        ca_b64=base64.b64encode(ca_pem.encode()).decode()))
+    clean_env.setenv("KUBECONFIG", str(kc))
+    cfg = core.resolve_kube_config()
+    assert cfg["url"] == "https://api.prod.example:6443"
+    assert cfg["token"] == "kubeconfig-token-123"
+    assert cfg["default_namespace"] == "ml-team"
+    # -data CA decoded to a temp file
+    assert cfg["ca_file"] and open(cfg["ca_file"]).read() == ca_pem
+
+
+def test_kubeconfig_insecure_and_cert_files(core, clean_env, tmp_path):
+    kc = tmp_path / "config"
+    kc.write_text("""\
+current-context: c
+clusters:
+- name: cl
+  cluster:
+    server: https://x:6443
+    insecure-skip-tls-verify: true
+contexts:
+- name: c
+  context:
+    cluster: cl
+    user: u
+users:
+- name: u
+  user:
+    client-certificate: /etc/certs/me.crt
+    client-key: /etc/certs/me.key
+""")
+    clean_env.setenv("KUBECONFIG", str(kc))
+    cfg = core.resolve_kube_config()
+    assert cfg["skip_tls"] is True
+    # client cert/key paths surface through KubeClient → http mTLS options
+
+
+def test_env_override_beats_kubeconfig(core, clean_env, tmp_path):
+    kc = tmp_path / "config"
+    kc.write_text("current-context: c\n")
+    clean_env.setenv("KUBECONFIG", str(kc))
+    clean_env.setenv("GPU_PRUNER_K8S_URL", "http://127.0.0.1:1234")
+    assert core.resolve_kube_config()["url"] == "http://127.0.0.1:1234"
+
+
+def test_prom_token_from_kubeconfig(core, clean_env, tmp_path):
+    import base64
+
+    kc = tmp_path / "config"
+    kc.write_text(KUBECONFIG_TMPL.format(
+        ca_b64=base64.b64encode(b"x").decode()))
+    clean_env.setenv("KUBECONFIG", str(kc))
+    clean_env.setenv("GPU_PRUNER_SA_DIR", str(tmp_path / "nosa"))
+    assert core.get_prometheus_token() == "kubeconfig-token-123"

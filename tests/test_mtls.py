@@ -90,3 +90,51 @@ def test_mtls_without_client_cert_rejected(pruner_bin, mtls_api, fake_prom, pki)
     r = run_pruner(pruner_bin, mtls_api.url, fake_prom.url, pki, with_client_cert=False)
     assert mtls_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 1
     assert "TLS" in r.stderr or "error" in r.stderr.lower()
+
+
+def test_kubeconfig_with_inline_mtls_material(pruner_bin, mtls_api, fake_prom, pki,
+                                              tmp_path, monkeypatch):
+    """A kubectl-style kubeconfig with base64 -data CA + client cert/key
+    drives the full cull against an mTLS apiserver — no env overrides."""
+    import base64
+
+    b64 = lambda p: base64.b64encode(open(p, "rb").read()).decode()
+    kc = tmp_path / "kubeconfig"
+    kc.write_text(f"""\
+apiVersion: v1
+kind: Config
+current-context: test
+clusters:
+- name: test-cluster
+  cluster:
+    server: {mtls_api.url}
+    certificate-authority-data: {b64(pki["ca"])}
+contexts:
+- name: test
+  context:
+    cluster: test-cluster
+    user: test-user
+users:
+- name: test-user
+  user:
+    client-certificate-data: {b64(pki["client_crt"])}
+    client-key-data: {b64(pki["client_key"])}
+""")
+    dep = mtls_api.add_deployment("kc-dep", "ml")
+    rs = mtls_api.add_replicaset("kc-dep-rs", "ml", owner=dep)
+    mtls_api.add_pod("kc-0", "ml", owner_kind="ReplicaSet", owner_name="kc-dep-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("kc-0", "ml")
+
+    env = dict(os.environ)
+    for var in ("GPU_PRUNER_K8S_URL", "GPU_PRUNER_K8S_CA",
+                "GPU_PRUNER_K8S_CLIENT_CERT", "GPU_PRUNER_K8S_CLIENT_KEY",
+                "KUBERNETES_SERVICE_HOST"):
+        env.pop(var, None)
+    env["KUBECONFIG"] = str(kc)
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--run-mode", "scale-down"],
+        capture_output=True, text=True, timeout=30, env=env)
+    assert r.returncode == 0, r.stderr
+    assert mtls_api.get("Deployment", "ml", "kc-dep")["spec"]["replicas"] == 0
