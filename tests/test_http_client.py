@@ -147,3 +147,42 @@ def test_server_survives_malformed_requests(core):
         assert r.status == 200
     finally:
         b.stop()
+
+
+def test_ipv6_literal_url(core):
+    """http://[::1]:port/... parses and connects."""
+    import http.server
+    import socket
+    import socketserver
+    import threading
+
+    class V6Server(socketserver.ThreadingTCPServer):
+        address_family = socket.AF_INET6
+        allow_reuse_address = True
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+
+        def log_message(self, *args):
+            pass
+
+        def do_GET(self):
+            body = b"v6-ok"
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+    try:
+        srv = V6Server(("::1", 0), Handler)
+    except OSError:
+        pytest.skip("no IPv6 loopback in this environment")
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        port = srv.server_address[1]
+        r = core._http_get(f"http://[::1]:{port}/x")
+        assert r["status"] == 200 and r["body"] == b"v6-ok"
+    finally:
+        srv.shutdown()
+        srv.server_close()

@@ -104,3 +104,11 @@ def test_log_level_filter_env(pruner_bin, idle_cluster, fake_prom):
     r2 = run_daemon(pruner_bin, idle_cluster, fake_prom,
                     env_extra={"RUST_LOG": "error", "GPU_PRUNER_LOG": ""})
     assert not any(" INFO " in l for l in r2.stderr.splitlines())
+
+
+def test_log_format_pretty(pruner_bin, idle_cluster, fake_prom):
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom, "--log-format", "pretty")
+    assert r.returncode == 0
+    # pretty format: level on its own styled segment, message indented
+    assert any(l.startswith("    ") for l in r.stderr.splitlines())
+    assert " INFO " in r.stderr
