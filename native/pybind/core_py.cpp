@@ -20,6 +20,7 @@
 #include "../pruner/promql.hpp"
 #include "../pruner/resources.hpp"
 #include "../common/http.hpp"
+#include "../common/miniyaml.hpp"
 #include "../pruner/otlp.hpp"
 #include "../pruner/synthbench.hpp"
 
@@ -320,6 +321,12 @@ PYBIND11_MODULE(_pruner_core, m) {
           d["headers"] = headers;
           return d;
         });
+
+  // test helper: the kubeconfig YAML-subset reader (common/miniyaml.hpp)
+  m.def("_yaml_to_json", [](const std::string& src) {
+    return miniyaml::parse(src).dump();
+  });
+  py::register_exception<miniyaml::Error>(m, "YamlError");
 
   m.def("counters_snapshot", [] {
     py::dict d;
