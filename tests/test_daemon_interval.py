@@ -147,3 +147,30 @@ def test_five_thousand_pod_tick_correctness(core, monkeypatch):
         assert b.events_posted == 2500
     finally:
         b.stop()
+
+
+def test_latency_injected_tick_uses_concurrency(core, monkeypatch):
+    """Catastrophic-concurrency-regression guard: with 2 ms injected RTT and
+    200 pods, a serial engine would need >= 200 x 3 x 2 ms ~ 1.2 s per tick;
+    the concurrent engine must finish far faster (very loose bound to stay
+    flake-free in CI)."""
+    import json as _json
+    import time as _time
+
+    monkeypatch.setenv("PROMETHEUS_TOKEN", "t")
+    b = core.SyntheticBackend(n_pods=200, latency_us=2000)
+    b.start()
+    try:
+        monkeypatch.setenv("GPU_PRUNER_K8S_URL", b.k8s_url)
+        cfg = _json.dumps({"duration": 30, "grace_period": 300,
+                           "run_mode": "scale-down",
+                           "prometheus_url": b.prom_url, "max_concurrency": 32,
+                           "eval_strategy": "get"})
+        core.run_tick(cfg)  # warmup
+        t0 = _time.perf_counter()
+        out = core.run_tick(cfg)
+        dt = _time.perf_counter() - t0
+        assert out["num_unique_pods"] == 200
+        assert dt < 0.8, f"tick took {dt:.2f}s — concurrency regression?"
+    finally:
+        b.stop()
