@@ -50,7 +50,14 @@ std::optional<ScaleKind> find_root_object(ObjectCache& objs, const jsn::Value& p
       std::string owner_name = owner.get("name").as_string();
       if (owner_kind == "ReplicaSet") {
         LOGI(TARGET, "Found ReplicaSet!");
-        auto rs = objs.get_object(Kind::ReplicaSet, ns, owner_name);
+        // fetch errors fall through to the next owner reference (the
+        // reference's `if let Ok(rs)` swallows them — lib.rs:464)
+        std::optional<jsn::Value> rs;
+        try {
+          rs = objs.get_object(Kind::ReplicaSet, ns, owner_name);
+        } catch (const std::exception& e) {
+          LOGW(TARGET, "ReplicaSet " + owner_name + " fetch failed: " + e.what());
+        }
         if (rs) {
           const jsn::Value& rs_ors = rs->at({"metadata", "ownerReferences"});
           if (rs_ors.is_array()) {
@@ -68,7 +75,12 @@ std::optional<ScaleKind> find_root_object(ObjectCache& objs, const jsn::Value& p
         }
       } else if (owner_kind == "StatefulSet") {
         LOGI(TARGET, "Found StatefulSet!");
-        auto ss = objs.get_object(Kind::StatefulSet, ns, owner_name);
+        std::optional<jsn::Value> ss;
+        try {
+          ss = objs.get_object(Kind::StatefulSet, ns, owner_name);
+        } catch (const std::exception& e) {
+          LOGW(TARGET, "StatefulSet " + owner_name + " fetch failed: " + e.what());
+        }
         if (ss) {
           const jsn::Value& ss_ors = ss->at({"metadata", "ownerReferences"});
           if (ss_ors.is_array()) {

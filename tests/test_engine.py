@@ -297,3 +297,38 @@ def test_list_strategy_uses_few_requests(core, fake_api):
     n_get = len(fake_api.requests)
     assert n_list <= 8, f"LIST strategy made {n_list} requests"
     assert n_get >= 40, f"GET strategy made {n_get} requests"
+
+
+def test_kserve_label_missing_isvc_skips(core, fake_api):
+    """KServe label pointing at a deleted InferenceService: skip, no crash
+    (reference propagates the error → pod skipped)."""
+    fake_api.add_pod("stale-pred", "ml",
+                     labels={"serving.kserve.io/inferenceservice": "gone"})
+    out = evaluate(core, [series("stale-pred", "ml")])
+    assert out["shutdown_events"] == 0
+
+
+def test_garbage_creation_timestamp_skipped(core, fake_api):
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    pod = fake_api.add_pod("badts", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                           owner_uid=rs["metadata"]["uid"])
+    pod["metadata"]["creationTimestamp"] = "not-a-date"
+    out = evaluate(core, [series("badts", "ml")])
+    assert out["shutdown_events"] == 0
+
+
+def test_owner_fetch_error_falls_through_to_next_owner(core, fake_api):
+    """A failing (500) owner GET tries the next ownerReference (reference
+    lib.rs:464 `if let Ok(rs)` semantics)."""
+    ss = fake_api.add_statefulset("db", "ml")
+    pod = fake_api.add_pod("multi", "ml")
+    pod["metadata"]["ownerReferences"] = [
+        # first owner: a ReplicaSet whose name breaks the fake's routing →
+        # the GET raises; the walk must continue to the StatefulSet
+        {"apiVersion": "apps/v1", "kind": "ReplicaSet", "name": "x/y", "uid": "u1"},
+        {"apiVersion": "apps/v1", "kind": "StatefulSet", "name": "db",
+         "uid": ss["metadata"]["uid"]},
+    ]
+    out = evaluate(core, [series("multi", "ml")])
+    assert [r.kind for r in out["roots"]] == ["StatefulSet"]
