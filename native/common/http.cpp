@@ -256,15 +256,28 @@ private:
     fd_ = fd;
   }
 
+  static bool is_ip_literal(const std::string& host) {
+    unsigned char buf[sizeof(struct in6_addr)];
+    return inet_pton(AF_INET, host.c_str(), buf) == 1 ||
+           inet_pton(AF_INET6, host.c_str(), buf) == 1;
+  }
+
   void start_tls(SSL_CTX* ctx, const std::string& host) {
     ensure_ssl_init();
     ssl_ = SSL_new(ctx);
     if (!ssl_) throw Error("SSL_new failed");
     SSL_set_fd(ssl_, fd_);
-    SSL_set_tlsext_host_name(ssl_, host.c_str());
-    // hostname verification (when verify enabled on the ctx)
+    // Peer identity check (when verify enabled on the ctx). An IP-literal host
+    // (the in-cluster apiserver is https://$KUBERNETES_SERVICE_HOST — a
+    // ClusterIP) must be matched against IP SANs, not DNS names, and RFC 6066
+    // forbids SNI for IP literals.
     X509_VERIFY_PARAM* param = SSL_get0_param(ssl_);
-    X509_VERIFY_PARAM_set1_host(param, host.c_str(), 0);
+    if (is_ip_literal(host)) {
+      X509_VERIFY_PARAM_set1_ip_asc(param, host.c_str());
+    } else {
+      SSL_set_tlsext_host_name(ssl_, host.c_str());
+      X509_VERIFY_PARAM_set1_host(param, host.c_str(), 0);
+    }
     while (true) {
       int rc = SSL_connect(ssl_);
       if (rc == 1) break;

@@ -38,8 +38,13 @@ def pki(tmp_path_factory):
         return key, crt
 
     s_key, s_crt = issue("server", "127.0.0.1", san="IP:127.0.0.1")
+    # Realistic apiserver cert shape: CN is NOT the IP, identity lives only in
+    # the IP SAN (in-cluster apiserver certs look like this; a client that
+    # runs only a DNS-name check fails the handshake against it).
+    ip_key, ip_crt = issue("server-ipsan", "kube-apiserver", san="IP:127.0.0.1")
     c_key, c_crt = issue("client", "gpu-pruner-user")
     return {"ca": str(ca_crt), "server_key": str(s_key), "server_crt": str(s_crt),
+            "ipsan_key": str(ip_key), "ipsan_crt": str(ip_crt),
             "client_key": str(c_key), "client_crt": str(c_crt)}
 
 
@@ -90,6 +95,26 @@ def test_mtls_without_client_cert_rejected(pruner_bin, mtls_api, fake_prom, pki)
     r = run_pruner(pruner_bin, mtls_api.url, fake_prom.url, pki, with_client_cert=False)
     assert mtls_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 1
     assert "TLS" in r.stderr or "error" in r.stderr.lower()
+
+
+def test_ip_literal_host_verified_against_ip_san(pruner_bin, fake_prom, pki):
+    """In-cluster operation: the apiserver URL is an IP literal (ClusterIP) and
+    the serving cert carries the identity only in an IP SAN (CN is the
+    component name, not the address). Verification must match the IP SAN —
+    a DNS-name-only check fails this handshake fail-closed
+    (native/common/http.cpp start_tls)."""
+    from gpu_pruner_amd.fixtures import FakeApiServer
+
+    with FakeApiServer(certfile=pki["ipsan_crt"], keyfile=pki["ipsan_key"]) as api:
+        assert api.url.startswith("https://127.0.0.1:")
+        dep = api.add_deployment("d", "ml")
+        rs = api.add_replicaset("d-rs", "ml", owner=dep)
+        api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                    owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+        fake_prom.add_idle_series("p0", "ml")
+        r = run_pruner(pruner_bin, api.url, fake_prom.url, pki, with_client_cert=False)
+        assert r.returncode == 0, r.stderr
+        assert api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0
 
 
 def test_kubeconfig_with_inline_mtls_material(pruner_bin, mtls_api, fake_prom, pki,
