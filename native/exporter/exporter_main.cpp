@@ -41,6 +41,9 @@ OPTIONS:
   -i, --interval <MS>      activity poll interval [default: 1000]
       --idle-epsilon <R>   windowed activity ratios below R report as exactly
                            0 (firmware housekeeping noise floor) [default: 0.005]
+      --activity-window <S> sliding window (seconds) the activity ratio is
+                           computed over; scrapes are idempotent — concurrent
+                           scrapers all observe the same window [default: 30]
       --node-type <STR>    value for the node_type const label
       --hostname <STR>     override Hostname label (default: gethostname)
   -l, --log-format <FMT>   default | json | pretty
@@ -70,6 +73,7 @@ int main(int argc, char** argv) {
   std::string bind_addr = "0.0.0.0";
   int interval_ms = 1000;
   double idle_epsilon = 0.005;
+  double activity_window_s = 30.0;
   std::string node_type;
   std::string hostname;
   logx::Format fmt = logx::Format::Default;
@@ -87,6 +91,7 @@ int main(int argc, char** argv) {
     else if (a == "-b" || a == "--bind") bind_addr = next();
     else if (a == "-i" || a == "--interval") interval_ms = std::stoi(next());
     else if (a == "--idle-epsilon") idle_epsilon = std::stod(next());
+    else if (a == "--activity-window") activity_window_s = std::stod(next());
     else if (a == "--node-type") node_type = next();
     else if (a == "--hostname") hostname = next();
     else if (a == "-l" || a == "--log-format") {
@@ -113,7 +118,7 @@ int main(int argc, char** argv) {
     else if (::gethostname(buf, sizeof buf - 1) == 0) hostname = buf;
   }
 
-  exporter::Sampler sampler(interval_ms, idle_epsilon);
+  exporter::Sampler sampler(interval_ms, idle_epsilon, activity_window_s);
   try {
     sampler.init();
   } catch (const std::exception& e) {
@@ -130,7 +135,10 @@ int main(int argc, char** argv) {
   http::Server server(bind_addr, port, [&](const http::ServerRequest& req) {
     http::ServerResponse resp;
     if (req.path == "/metrics") {
-      auto samples = sampler.snapshot(/*reset_window=*/true);
+      // Idempotent scrape: snapshot() is read-only (fixed sliding window),
+      // so concurrent scrapers (HA Prometheus pairs, debug curls) cannot
+      // truncate the activity window another scraper observes.
+      auto samples = sampler.snapshot();
       auto attribs = attributor.resolve_full(samples);
       resp.body = exporter::render_metrics(samples, attribs, opts);
       resp.content_type = "text/plain; version=0.0.4; charset=utf-8";

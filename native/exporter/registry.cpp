@@ -73,12 +73,28 @@ std::string render_metrics(const std::vector<DeviceSample>& samples,
        "counter", [](const DeviceSample& d) { return d.xgmi_write_kb; }},
   };
 
+  // Sampler read-health families (always emitted, healthy or not): alerting
+  // surface for a device whose SMU/sysfs reads are failing.
+  static const Family health_families[] = {
+      {"mi355_sampler_healthy",
+       "1 when device activity reads are succeeding; 0 after repeated failures "
+       "(activity series are withheld while 0)",
+       "gauge", [](const DeviceSample& d) { return d.healthy ? 1.0 : 0.0; }},
+      {"mi355_sampler_last_good_read_age_seconds",
+       "Seconds since the last successful activity read for this device",
+       "gauge", [](const DeviceSample& d) { return d.staleness_s; }},
+  };
+
   std::string out;
   out.reserve(4096);
-  for (const auto& fam : families) {
+  auto emit_family = [&](const Family& fam, bool healthy_only) {
     out += std::string("# HELP ") + fam.name + " " + fam.help + "\n";
     out += std::string("# TYPE ") + fam.name + " " + fam.type + "\n";
     for (const auto& d : samples) {
+      // A device with a dead read path must not publish (stale) values:
+      // a withheld series can never satisfy the culler's `== 0` idle
+      // predicate, so broken telemetry fails safe (no cull).
+      if (healthy_only && !d.healthy) continue;
       out += fam.name;
       out += "{gpu=\"" + std::to_string(d.index) + "\"";
       out += ",UUID=\"" + escape_label(d.unique_id) + "\"";
@@ -94,7 +110,9 @@ std::string render_metrics(const std::vector<DeviceSample>& samples,
         out += "," + k + "=\"" + escape_label(v) + "\"";
       out += "} " + fmt(fam.get(d)) + "\n";
     }
-  }
+  };
+  for (const auto& fam : families) emit_family(fam, /*healthy_only=*/true);
+  for (const auto& fam : health_families) emit_family(fam, /*healthy_only=*/false);
   return out;
 }
 

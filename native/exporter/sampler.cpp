@@ -56,8 +56,8 @@ uint64_t kfd_gpu_id_for_bdf(uint64_t bdfid) {
 
 }  // namespace
 
-Sampler::Sampler(int poll_interval_ms, double idle_epsilon)
-    : poll_interval_ms_(poll_interval_ms), idle_epsilon_(idle_epsilon) {}
+Sampler::Sampler(int poll_interval_ms, double idle_epsilon, double window_s)
+    : poll_interval_ms_(poll_interval_ms), idle_epsilon_(idle_epsilon), window_s_(window_s) {}
 
 Sampler::~Sampler() {
   stop();
@@ -77,6 +77,7 @@ void Sampler::init() {
 
   devices_.resize(n_devices_);
   for (uint32_t i = 0; i < n_devices_; i++) {
+    devices_[i].win.set_retention(window_s_ * 2.0 + 10.0);
     DeviceSample& d = devices_[i].last;
     d.index = i;
     char name[256] = {0};
@@ -150,11 +151,12 @@ void Sampler::poll_device(uint32_t i) {
     std::lock_guard<std::mutex> lock(mu_);
     d = devices_[i].last;
   }
+  d.metrics_table_ok = false;
   double now = monotonic_s();
 
   uint32_t busy = 0;
-  if (rsmi_dev_busy_percent_get(i, &busy) == RSMI_STATUS_SUCCESS)
-    d.busy_percent = static_cast<double>(busy);
+  bool busy_read_ok = rsmi_dev_busy_percent_get(i, &busy) == RSMI_STATUS_SUCCESS;
+  if (busy_read_ok) d.busy_percent = static_cast<double>(busy);
 
   uint32_t mem_busy = 0;
   if (rsmi_dev_memory_busy_percent_get(i, &mem_busy) == RSMI_STATUS_SUCCESS)
@@ -201,59 +203,75 @@ void Sampler::poll_device(uint32_t i) {
       d.busy_percent = gm.average_gfx_activity;
   }
 
-  // ---- integrate the scrape-window activity ratio (brief state lock) ----
+  // An activity observation exists when either the busy-percent sysfs read
+  // or the firmware gpu_metrics table succeeded. When BOTH fail the span
+  // since the previous poll is recorded as UNKNOWN — it never contributes
+  // idle time — and the device's health counter advances (a
+  // dying SMU read path must not make a busy GPU look permanently idle).
+  bool activity_ok = busy_read_ok || d.metrics_table_ok;
+  d.read_ok = activity_ok;
+
+  // ---- append the poll segment to the sliding window (brief state lock) ----
   std::lock_guard<std::mutex> lock(mu_);
   DevState& st = devices_[i];
-  if (st.prev_poll_monotonic > 0.0) {
-    double dt = now - st.prev_poll_monotonic;
-    if (dt > 0) {
-      double ratio = d.busy_percent / 100.0;
-      // When the firmware accumulator advanced, derive the exact ratio over
-      // the poll interval from Δacc/Δt. Units calibrated on MI355X silicon
-      // (profiles/raw/calibration.log): firmware_timestamp ticks in ns on
-      // gfx950 (the rocm_smi header documents 10 ns — we auto-detect by
-      // comparing against the host monotonic delta), and gfx_activity_acc
-      // accumulates 100,000 counts per second at 100% busy.
-      if (d.metrics_table_ok && st.have_prev_acc &&
-          d.firmware_timestamp > st.prev_fw_ts) {
-        double fw_dt = static_cast<double>(d.firmware_timestamp - st.prev_fw_ts);
-        double fw_dt_ns = fw_dt * 1e-9, fw_dt_10ns = fw_dt * 1e-8;
-        double fw_dt_s =
-            std::abs(fw_dt_ns - dt) <= std::abs(fw_dt_10ns - dt) ? fw_dt_ns : fw_dt_10ns;
-        constexpr double kAccFullRate = 100000.0;  // counts/s at 100% (measured)
-        if (fw_dt_s > 0) {
-          double acc_ratio = static_cast<double>(d.gfx_activity_acc - st.prev_acc) /
-                             (fw_dt_s * kAccFullRate);
-          if (acc_ratio >= 0.0 && acc_ratio <= 1.5) ratio = std::min(acc_ratio, 1.0);
-        }
+  {
+    double ratio = d.busy_percent / 100.0;
+    // When the firmware accumulator advanced, derive the exact ratio over
+    // the poll interval from Δacc/Δt. Units calibrated on MI355X silicon
+    // (profiles/raw/calibration.log): firmware_timestamp ticks in ns on
+    // gfx950 (the rocm_smi header documents 10 ns — we auto-detect by
+    // comparing against the host monotonic delta), and gfx_activity_acc
+    // accumulates 100,000 counts per second at 100% busy.
+    double dt = now - st.win_prev_t;
+    if (d.metrics_table_ok && st.have_prev_acc && st.win_prev_t > 0.0 && dt > 0 &&
+        d.firmware_timestamp > st.prev_fw_ts) {
+      double fw_dt = static_cast<double>(d.firmware_timestamp - st.prev_fw_ts);
+      double fw_dt_ns = fw_dt * 1e-9, fw_dt_10ns = fw_dt * 1e-8;
+      double fw_dt_s =
+          std::abs(fw_dt_ns - dt) <= std::abs(fw_dt_10ns - dt) ? fw_dt_ns : fw_dt_10ns;
+      constexpr double kAccFullRate = 100000.0;  // counts/s at 100% (measured)
+      if (fw_dt_s > 0) {
+        double acc_ratio = static_cast<double>(d.gfx_activity_acc - st.prev_acc) /
+                           (fw_dt_s * kAccFullRate);
+        if (acc_ratio >= 0.0 && acc_ratio <= 1.5) ratio = std::min(acc_ratio, 1.0);
       }
-      st.busy_seconds += ratio * dt;
-      st.wall_seconds += dt;
     }
+    st.win_prev_t = now;
+    st.win.add(now, ratio, activity_ok);
   }
-  st.prev_poll_monotonic = now;
   st.prev_acc = d.gfx_activity_acc;
   st.prev_fw_ts = d.firmware_timestamp;
   st.have_prev_acc = d.metrics_table_ok;
 
-  double window_ratio =
-      st.wall_seconds > 0 ? std::min(st.busy_seconds / st.wall_seconds, 1.0) : 0.0;
-  // Floor sub-noise ratios to an exact 0 (see ctor comment): the idle
-  // predicate is `== 0`, and housekeeping blips are not workload.
-  d.gr_engine_active = window_ratio < idle_epsilon_ ? 0.0 : window_ratio;
+  if (activity_ok) {
+    st.consecutive_failures = 0;
+    st.last_good_monotonic = now;
+  } else {
+    st.consecutive_failures++;
+    if (st.consecutive_failures == kUnhealthyAfter)
+      LOGE(TARGET, "device " + std::to_string(i) + ": " +
+                       std::to_string(st.consecutive_failures) +
+                       " consecutive failed activity reads — marking unhealthy "
+                       "(activity series will be withheld)");
+  }
   st.last = d;
 }
 
-std::vector<DeviceSample> Sampler::snapshot(bool reset_window) {
+std::vector<DeviceSample> Sampler::snapshot() {
   std::lock_guard<std::mutex> lock(mu_);
+  double now = monotonic_s();
   std::vector<DeviceSample> out;
   out.reserve(devices_.size());
   for (auto& st : devices_) {
-    out.push_back(st.last);
-    if (reset_window) {
-      st.busy_seconds = 0.0;
-      st.wall_seconds = 0.0;
-    }
+    DeviceSample d = st.last;
+    double known_s = 0.0;
+    double window_ratio = std::min(st.win.ratio(now, window_s_, &known_s), 1.0);
+    // Floor sub-noise ratios to an exact 0 (see ctor comment): the idle
+    // predicate is `== 0`, and housekeeping blips are not workload.
+    d.gr_engine_active = window_ratio < idle_epsilon_ ? 0.0 : window_ratio;
+    d.healthy = st.consecutive_failures < kUnhealthyAfter;
+    d.staleness_s = st.last_good_monotonic > 0.0 ? now - st.last_good_monotonic : 0.0;
+    out.push_back(std::move(d));
   }
   return out;
 }

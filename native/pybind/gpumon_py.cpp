@@ -40,6 +40,9 @@ py::dict sample_to_dict(const DeviceSample& d) {
   s["metrics_table_ok"] = d.metrics_table_ok;
   s["gfx_activity_acc"] = d.gfx_activity_acc;
   s["firmware_timestamp"] = d.firmware_timestamp;
+  s["read_ok"] = d.read_ok;
+  s["healthy"] = d.healthy;
+  s["staleness_s"] = d.staleness_s;
   s["xgmi_link_width"] = d.xgmi_link_width;
   s["xgmi_link_speed"] = d.xgmi_link_speed;
   s["xgmi_read_kb"] = d.xgmi_read_kb;
@@ -63,6 +66,8 @@ DeviceSample sample_from_json(const jsn::Value& v) {
   d.vram_total_b = v.get("vram_total_b").as_double(0);
   d.temp_edge_c = v.get("temp_edge_c").as_double(0);
   d.gfx_clock_mhz = v.get("gfx_clock_mhz").as_double(0);
+  d.healthy = v.get("healthy").as_bool(true);
+  d.staleness_s = v.get("staleness_s").as_double(0);
   return d;
 }
 
@@ -76,8 +81,8 @@ PYBIND11_MODULE(_gpumon, m) {
   py::register_exception<SamplerError>(m, "SamplerError");
 
   py::class_<Sampler>(m, "Sampler")
-      .def(py::init<int, double>(), py::arg("poll_interval_ms") = 1000,
-           py::arg("idle_epsilon") = 0.005)
+      .def(py::init<int, double, double>(), py::arg("poll_interval_ms") = 1000,
+           py::arg("idle_epsilon") = 0.005, py::arg("window_s") = 30.0)
       .def("init", &Sampler::init,
            "Initialize rocm_smi and enumerate devices (raises SamplerError "
            "without an AMD GPU)")
@@ -85,14 +90,29 @@ PYBIND11_MODULE(_gpumon, m) {
       .def("stop", &Sampler::stop)
       .def("poll_once", &Sampler::poll_once, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("device_count", &Sampler::device_count)
-      .def("snapshot",
-           [](Sampler& s, bool reset_window) {
-             auto samples = s.snapshot(reset_window);
-             py::list out;
-             for (const auto& d : samples) out.append(sample_to_dict(d));
-             return out;
+      .def("snapshot", [](Sampler& s) {
+        auto samples = s.snapshot();
+        py::list out;
+        for (const auto& d : samples) out.append(sample_to_dict(d));
+        return out;
+      });
+
+  // Pure sliding-window integrator — CPU-unit-testable core of the sampler's
+  // scrape-idempotent activity ratio (no rsmi involved).
+  py::class_<ActivityWindow>(m, "ActivityWindow")
+      .def(py::init<>())
+      .def("add", &ActivityWindow::add, py::arg("t"), py::arg("ratio"),
+           py::arg("known") = true)
+      .def("ratio",
+           [](const ActivityWindow& w, double now, double window_s) {
+             double known = 0.0;
+             double r = w.ratio(now, window_s, &known);
+             return py::make_tuple(r, known);
            },
-           py::arg("reset_window") = false);
+           py::arg("now"), py::arg("window_s"),
+           "-> (time-weighted ratio over known segments, known seconds)")
+      .def("set_retention", &ActivityWindow::set_retention)
+      .def_property_readonly("size", &ActivityWindow::size);
 
   m.def("list_pod_resources",
         [](const std::string& socket_path, int timeout_ms) {

@@ -87,6 +87,110 @@ def test_render_label_escaping(gpumon):
     assert 'modelName="AMD \\"MI355X\\"\\\\test"' in text
 
 
+# ---- sampler read-health -----------------------------------
+
+
+def test_render_unhealthy_device_withholds_activity_series(gpumon):
+    """A device whose SMU reads are failing must NOT publish (stale) DCGM
+    series — a withheld series can never satisfy the pruner's == 0 idle
+    predicate, so broken telemetry fails safe instead of culling."""
+    sick = dict(SAMPLE, healthy=False, staleness_s=42.5)
+    text = gpumon.render_metrics(json.dumps([sick]))
+    assert "DCGM_FI_PROF_GR_ENGINE_ACTIVE{" not in text
+    assert "DCGM_FI_DEV_GPU_UTIL{" not in text
+    # ...but the health families still report it for alerting
+    assert 'mi355_sampler_healthy{gpu="0"' in text
+    line = [l for l in text.splitlines() if l.startswith("mi355_sampler_healthy{")][0]
+    assert line.endswith("} 0")
+    age = [l for l in text.splitlines()
+           if l.startswith("mi355_sampler_last_good_read_age_seconds{")][0]
+    assert age.endswith("} 42.5")
+
+
+def test_render_healthy_device_has_health_series(gpumon):
+    text = gpumon.render_metrics(json.dumps([SAMPLE]))
+    line = [l for l in text.splitlines() if l.startswith("mi355_sampler_healthy{")][0]
+    assert line.endswith("} 1")
+    assert "DCGM_FI_PROF_GR_ENGINE_ACTIVE{" in text
+
+
+def test_render_mixed_health_per_device(gpumon):
+    samples = [dict(SAMPLE, index=0),
+               dict(SAMPLE, index=1, healthy=False, staleness_s=10.0)]
+    text = gpumon.render_metrics(json.dumps(samples))
+    util = [l for l in text.splitlines() if l.startswith("DCGM_FI_DEV_GPU_UTIL{")]
+    assert len(util) == 1 and 'gpu="0"' in util[0]
+    health = [l for l in text.splitlines() if l.startswith("mi355_sampler_healthy{")]
+    assert len(health) == 2
+
+
+# ---- ActivityWindow: scrape-idempotent sliding window -----------------------
+
+
+def test_activity_window_basic_ratio(gpumon):
+    w = gpumon.ActivityWindow()
+    for i in range(11):  # 10 x 1s segments, half busy
+        w.add(float(i), 1.0 if i % 2 == 0 else 0.0)
+    r, known = w.ratio(10.0, 10.0)
+    assert known == pytest.approx(10.0)
+    assert 0.3 < r < 0.7
+
+
+def test_activity_window_is_idempotent(gpumon):
+    w = gpumon.ActivityWindow()
+    for i in range(6):
+        w.add(float(i), 0.8)
+    first = w.ratio(5.0, 5.0)
+    second = w.ratio(5.0, 5.0)  # a second "scraper" sees the same value
+    assert first == second == (pytest.approx(0.8), pytest.approx(5.0))
+
+
+def test_activity_window_burst_ages_out(gpumon):
+    w = gpumon.ActivityWindow()
+    w.add(0.0, 0.0)
+    w.add(1.0, 1.0)   # busy during [0,1)
+    for t in range(2, 40):
+        w.add(float(t), 0.0)
+    r, _ = w.ratio(39.0, 5.0)  # window [34,39] is all idle
+    assert r == 0.0
+
+
+def test_activity_window_unknown_segments_never_count_as_idle(gpumon):
+    """Failed reads (known=False) contribute to neither busy nor known time:
+    a dead read path cannot decay the ratio toward a false 0."""
+    w = gpumon.ActivityWindow()
+    w.add(0.0, 0.0)
+    w.add(1.0, 1.0)  # known busy second
+    for t in range(2, 8):
+        w.add(float(t), 0.0, False)  # reads failing
+    r, known = w.ratio(7.0, 7.0)
+    assert known == pytest.approx(1.0)
+    assert r == pytest.approx(1.0)  # the only known time was busy
+
+
+def test_activity_window_no_data(gpumon):
+    w = gpumon.ActivityWindow()
+    r, known = w.ratio(100.0, 30.0)
+    assert r == 0.0 and known == 0.0
+
+
+def test_activity_window_retention_bounds_memory(gpumon):
+    w = gpumon.ActivityWindow()
+    w.set_retention(10.0)
+    for i in range(1000):
+        w.add(float(i), 0.5)
+    assert w.size < 20
+
+
+def test_activity_window_partial_segment_weighting(gpumon):
+    w = gpumon.ActivityWindow()
+    w.add(0.0, 0.0)
+    w.add(10.0, 1.0)  # one long busy segment [0,10)
+    r, known = w.ratio(10.0, 5.0)  # window covers only [5,10)
+    assert known == pytest.approx(5.0)
+    assert r == pytest.approx(1.0)
+
+
 # ---- cgroup → pod UID -------------------------------------------------------
 
 

@@ -33,7 +33,9 @@ def sampler():
     _require_gpu()
     from gpu_pruner_amd import _gpumon
 
-    s = _gpumon.Sampler(poll_interval_ms=100)
+    # 2 s sliding window: short enough that idle/busy transitions age out
+    # within a test, long enough to span many 100 ms polls
+    s = _gpumon.Sampler(poll_interval_ms=100, window_s=2.0)
     s.init()  # must not silently fall back — raises SamplerError if broken
     yield s
     s.stop()
@@ -65,19 +67,19 @@ def test_idle_utilization_is_exactly_zero(sampler):
     """The PromQL `== 0` predicate depends on a literal zero when idle."""
     busy = _settle_idle(sampler, seconds=10.0)
     assert busy == 0.0, f"idle GPU reports busy_percent={busy}"
-    # windowed ratio over a fresh idle window
-    sampler.snapshot(True)  # reset window
-    time.sleep(1.0)
-    sampler.poll_once()
+    # keep polling so the 2 s sliding window is entirely idle time
+    for _ in range(25):
+        time.sleep(0.1)
+        sampler.poll_once()
     ratio = sampler.snapshot()[0]["gr_engine_active"]
     assert ratio < 0.01, f"idle GR_ENGINE_ACTIVE={ratio}"
+    assert sampler.snapshot()[0]["healthy"] is True
 
 
 def test_busy_probe_raises_utilization(sampler):
     from gpu_pruner_amd import probe
 
     _settle_idle(sampler, seconds=5.0)
-    sampler.snapshot(True)
     probe.start(device=0, max_seconds=30.0)
     try:
         busy = 0.0
@@ -148,8 +150,20 @@ def test_exporter_binary_serves_real_metrics():
         for fam in ("DCGM_FI_DEV_FB_USED", "DCGM_FI_DEV_FB_FREE",
                     "DCGM_FI_DEV_GPU_TEMP", "DCGM_FI_DEV_SM_CLOCK",
                     "DCGM_FI_DEV_TOTAL_ENERGY_CONSUMPTION",
-                    "mi355_xgmi_link_width", "mi355_xgmi_read_kb_total"):
-            assert f"# TYPE {fam} " in text, f"missing family {fam}" 
+                    "mi355_xgmi_link_width", "mi355_xgmi_read_kb_total",
+                    "mi355_sampler_healthy",
+                    "mi355_sampler_last_good_read_age_seconds"):
+            assert f"# TYPE {fam} " in text, f"missing family {fam}"
+        healthy = _parse_prom_text(text, "mi355_sampler_healthy")
+        assert healthy and all(v == 1.0 for _, v in healthy), healthy
+        # scrapes are idempotent: two back-to-back scrapes agree on the
+        # windowed activity value (reset-on-scrape would zero the second)
+        text2 = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
+        a1 = _parse_prom_text(text, "DCGM_FI_PROF_GR_ENGINE_ACTIVE")
+        a2 = _parse_prom_text(text2, "DCGM_FI_PROF_GR_ENGINE_ACTIVE")
+        for (l1, v1), (l2, v2) in zip(a1, a2):
+            assert abs(v1 - v2) < 0.05, (v1, v2)
         health = urllib.request.urlopen(
             f"http://127.0.0.1:{port}/healthz", timeout=2).read()
         assert health == b"ok\n"
@@ -250,7 +264,7 @@ def test_busy_gpu_is_not_culled_idle_gpu_is():
     import json
     from gpu_pruner_amd import _gpumon, _pruner_core as core, probe
 
-    sampler = _gpumon.Sampler(poll_interval_ms=100)
+    sampler = _gpumon.Sampler(poll_interval_ms=100, window_s=1.0)
     sampler.init()
     backend = core.SyntheticBackend(n_pods=20)
     backend.start()
@@ -270,9 +284,10 @@ def test_busy_gpu_is_not_culled_idle_gpu_is():
                 busy = sampler.snapshot()[0]["busy_percent"]
                 if busy >= 90.0:
                     break
-            sampler.snapshot(True)
-            time.sleep(0.5)
-            sampler.poll_once()
+            # poll through a full 1 s window of busy time
+            for _ in range(12):
+                time.sleep(0.1)
+                sampler.poll_once()
             ratio = sampler.snapshot()[0]["gr_engine_active"]
             backend.set_series_value(ratio)
             assert ratio > 0.0
@@ -288,9 +303,10 @@ def test_busy_gpu_is_not_culled_idle_gpu_is():
             sampler.poll_once()
             if sampler.snapshot()[0]["busy_percent"] == 0.0:
                 break
-        sampler.snapshot(True)
-        time.sleep(0.5)
-        sampler.poll_once()
+        # poll through a full idle window so the busy burst ages out
+        for _ in range(15):
+            time.sleep(0.1)
+            sampler.poll_once()
         ratio = sampler.snapshot()[0]["gr_engine_active"]
         backend.set_series_value(ratio)
         assert ratio == 0.0, f"idle ratio={ratio}"
@@ -312,9 +328,11 @@ def test_window_semantics_with_real_gpu_signal(fake_api, pruner_bin):
     from gpu_pruner_amd import probe
 
     port = 19402
+    # short sliding window so the burst ages out of the *exporter's* window
+    # quickly; the lookback semantics under test live in MiniProm/PromQL
     exporter = subprocess.Popen(
         [str(REPO_ROOT / "bin" / "mi355-exporter"), "-p", str(port),
-         "-b", "127.0.0.1", "-i", "200"],
+         "-b", "127.0.0.1", "-i", "200", "--activity-window", "3"],
         stdout=subprocess.PIPE, stderr=subprocess.PIPE)
 
     def scrape_into(prom):
@@ -348,6 +366,7 @@ def test_window_semantics_with_real_gpu_signal(fake_api, pruner_bin):
             with probe.busy_load(device=0, max_seconds=20.0):
                 time.sleep(1.5)
                 busy_val = scrape_into(prom)
+            burst_t = time.monotonic()
             assert busy_val > 0.0
             # settle + scrape an idle sample
             deadline = time.monotonic() + 15
@@ -366,11 +385,11 @@ def test_window_semantics_with_real_gpu_signal(fake_api, pruner_bin):
             assert r.returncode == 0, r.stderr
             assert fake_api.get("Deployment", "ml", "train")["spec"]["replicas"] == 1
 
-            # wait for the burst to age out of the window, keep scraping idle
-            for _ in range(70):
+            # keep scraping idle samples until the burst has aged out of the
+            # 1-minute lookback (plus margin for the scrape-timestamp grid)
+            while time.monotonic() - burst_t < 75.0:
                 time.sleep(1.0)
-                if scrape_into(prom) != 0.0:
-                    continue
+                scrape_into(prom)
             r = subprocess.run(args, capture_output=True, text=True, timeout=30, env=env)
             assert r.returncode == 0, r.stderr
             assert fake_api.get("Deployment", "ml", "train")["spec"]["replicas"] == 0
