@@ -6,7 +6,9 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <openssl/err.h>
+#include <openssl/pem.h>
 #include <openssl/ssl.h>
+#include <openssl/x509.h>
 #include <poll.h>
 #include <sys/socket.h>
 #include <sys/un.h>
@@ -85,6 +87,49 @@ std::string ssl_err_string() {
   char buf[256];
   ERR_error_string_n(e, buf, sizeof buf);
   return buf;
+}
+
+// Add every certificate in an in-memory PEM bundle to the ctx's trust store.
+bool add_ca_pem(SSL_CTX* ctx, const std::string& pem) {
+  BIO* bio = BIO_new_mem_buf(pem.data(), static_cast<int>(pem.size()));
+  if (!bio) return false;
+  X509_STORE* store = SSL_CTX_get_cert_store(ctx);
+  int added = 0;
+  while (X509* cert = PEM_read_bio_X509(bio, nullptr, nullptr, nullptr)) {
+    if (X509_STORE_add_cert(store, cert) == 1) added++;
+    X509_free(cert);
+  }
+  ERR_clear_error();  // trailing-garbage error from the final read attempt
+  BIO_free(bio);
+  return added > 0;
+}
+
+// Load an in-memory client certificate chain + private key (kubeconfig
+// `-data` material — kept off the filesystem by design).
+bool use_client_pem(SSL_CTX* ctx, const std::string& cert_pem, const std::string& key_pem) {
+  BIO* cbio = BIO_new_mem_buf(cert_pem.data(), static_cast<int>(cert_pem.size()));
+  if (!cbio) return false;
+  X509* leaf = PEM_read_bio_X509(cbio, nullptr, nullptr, nullptr);
+  bool ok = leaf && SSL_CTX_use_certificate(ctx, leaf) == 1;
+  if (leaf) X509_free(leaf);
+  while (ok) {  // remaining certs in the bundle form the chain
+    X509* extra = PEM_read_bio_X509(cbio, nullptr, nullptr, nullptr);
+    if (!extra) break;
+    if (SSL_CTX_add_extra_chain_cert(ctx, extra) != 1) {  // ctx owns on success
+      X509_free(extra);
+      ok = false;
+    }
+  }
+  ERR_clear_error();
+  BIO_free(cbio);
+  if (!ok) return false;
+  BIO* kbio = BIO_new_mem_buf(key_pem.data(), static_cast<int>(key_pem.size()));
+  if (!kbio) return false;
+  EVP_PKEY* key = PEM_read_bio_PrivateKey(kbio, nullptr, nullptr, nullptr);
+  ok = key && SSL_CTX_use_PrivateKey(ctx, key) == 1;
+  if (key) EVP_PKEY_free(key);
+  BIO_free(kbio);
+  return ok && SSL_CTX_check_private_key(ctx) == 1;
 }
 
 }  // namespace
@@ -319,9 +364,14 @@ Client::Client(Url base, ClientOptions opts) : base_(std::move(base)), opts_(std
       case TlsVerify::CustomCa:
         SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER, nullptr);
         SSL_CTX_set_default_verify_paths(ctx);
-        if (SSL_CTX_load_verify_locations(ctx, opts_.ca_file.c_str(), nullptr) != 1) {
+        if (!opts_.ca_file.empty() &&
+            SSL_CTX_load_verify_locations(ctx, opts_.ca_file.c_str(), nullptr) != 1) {
           SSL_CTX_free(ctx);
           throw Error("failed to load CA bundle " + opts_.ca_file + ": " + ssl_err_string());
+        }
+        if (!opts_.ca_pem.empty() && !add_ca_pem(ctx, opts_.ca_pem)) {
+          SSL_CTX_free(ctx);
+          throw Error("failed to load in-memory CA bundle: " + ssl_err_string());
         }
         break;
     }
@@ -334,6 +384,11 @@ Client::Client(Url base, ClientOptions opts) : base_(std::move(base)), opts_(std
         SSL_CTX_free(ctx);
         throw Error("failed to load client certificate/key (" + opts_.client_cert_file +
                     ", " + opts_.client_key_file + "): " + ssl_err_string());
+      }
+    } else if (!opts_.client_cert_pem.empty()) {
+      if (!use_client_pem(ctx, opts_.client_cert_pem, opts_.client_key_pem)) {
+        SSL_CTX_free(ctx);
+        throw Error("failed to load in-memory client certificate/key: " + ssl_err_string());
       }
     }
     ssl_ctx_ = ctx;

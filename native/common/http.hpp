@@ -54,6 +54,11 @@ struct ClientOptions {
   std::string ca_file;          // PEM bundle for TlsVerify::CustomCa
   std::string client_cert_file; // mTLS client certificate (PEM), optional
   std::string client_key_file;  // mTLS client private key (PEM)
+  // In-memory PEM alternatives (kubeconfig base64 `-data` material): loaded
+  // via BIOs so key material never touches the filesystem.
+  std::string ca_pem;           // additional trusted roots for CustomCa
+  std::string client_cert_pem;  // mTLS client certificate chain
+  std::string client_key_pem;   // mTLS client private key
   int connect_timeout_ms = 5000;
   int io_timeout_ms = 30000;    // per-request read/write deadline
   int max_pool_per_origin = 256;

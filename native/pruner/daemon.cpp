@@ -162,7 +162,10 @@ int run_daemon(const Config& cfg) {
     } catch (const std::exception& e) {
       logx::counter_add("monotonic_counter.query_failures", 1);
       LOGE(TARGET, std::string("Failed to run query and scale down: ") + e.what());
-      if (++consecutive_failures > cfg.max_consecutive_failures) {
+      // Pre-increment comparison for reference parity (main.rs:310-321 reads
+      // the counter before bumping it): with the default max of 5 the daemon
+      // survives 6 consecutive failures and exits on the 7th.
+      if (consecutive_failures++ > cfg.max_consecutive_failures) {
         LOGE(TARGET, "Too many consecutive failures, exiting");
         exit_code.store(1);
         break;

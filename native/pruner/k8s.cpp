@@ -25,20 +25,6 @@ const char* env(const char* name) {
   return v && *v ? v : nullptr;
 }
 
-// Write decoded -data material to a private temp file; returns the path.
-std::string write_temp_pem(const std::string& data, const char* tag) {
-  std::string tmpl = std::string("/tmp/gpu-pruner-") + tag + "-XXXXXX";
-  std::vector<char> buf(tmpl.begin(), tmpl.end());
-  buf.push_back('\0');
-  int fd = ::mkstemp(buf.data());
-  if (fd < 0) throw std::runtime_error("mkstemp failed for kubeconfig material");
-  ssize_t n = ::write(fd, data.data(), data.size());
-  ::close(fd);
-  if (n != static_cast<ssize_t>(data.size()))
-    throw std::runtime_error("short write of kubeconfig material");
-  return std::string(buf.data());
-}
-
 // Minimal kubeconfig loader (current-context → cluster + user). The
 // reference gets this via kube-rs Config::infer; here a YAML-subset parser
 // (common/miniyaml.hpp) covers kubectl-generated files: server, CA
@@ -83,7 +69,7 @@ std::optional<KubeConfig> load_kubeconfig(const std::string& path) {
   else if (cluster.get("certificate-authority-data").is_string()) {
     std::string pem;
     if (miniyaml::base64_decode(cluster.get("certificate-authority-data").as_string(), &pem))
-      cfg.ca_file = write_temp_pem(pem, "ca");
+      cfg.ca_data = pem;
   }
   if (user.is_object()) {
     if (user.get("token").is_string()) cfg.token = user.get("token").as_string();
@@ -93,14 +79,14 @@ std::optional<KubeConfig> load_kubeconfig(const std::string& path) {
     else if (user.get("client-certificate-data").is_string()) {
       std::string pem;
       if (miniyaml::base64_decode(user.get("client-certificate-data").as_string(), &pem))
-        cfg.client_cert_file = write_temp_pem(pem, "cert");
+        cfg.client_cert_data = pem;
     }
     if (user.get("client-key").is_string())
       cfg.client_key_file = user.get("client-key").as_string();
     else if (user.get("client-key-data").is_string()) {
       std::string pem;
       if (miniyaml::base64_decode(user.get("client-key-data").as_string(), &pem))
-        cfg.client_key_file = write_temp_pem(pem, "key");
+        cfg.client_key_data = pem;
     }
   }
   if (ctx.get("namespace").is_string())
@@ -161,13 +147,17 @@ KubeClient::KubeClient(KubeConfig cfg) : cfg_(std::move(cfg)) {
   http::ClientOptions opts;
   if (cfg_.skip_tls) {
     opts.tls = http::TlsVerify::Skip;
-  } else if (cfg_.ca_file) {
+  } else if (cfg_.ca_file || cfg_.ca_data) {
     opts.tls = http::TlsVerify::CustomCa;
-    opts.ca_file = *cfg_.ca_file;
+    if (cfg_.ca_file) opts.ca_file = *cfg_.ca_file;
+    if (cfg_.ca_data) opts.ca_pem = *cfg_.ca_data;
   }
   if (cfg_.client_cert_file && cfg_.client_key_file) {
     opts.client_cert_file = *cfg_.client_cert_file;
     opts.client_key_file = *cfg_.client_key_file;
+  } else if (cfg_.client_cert_data && cfg_.client_key_data) {
+    opts.client_cert_pem = *cfg_.client_cert_data;
+    opts.client_key_pem = *cfg_.client_key_data;
   }
   http_ = std::make_unique<http::Client>(*url, opts);
 }

@@ -32,6 +32,12 @@ struct KubeConfig {
   // mTLS client-certificate auth (kube client-cert users); both required
   std::optional<std::string> client_cert_file;
   std::optional<std::string> client_key_file;
+  // Decoded kubeconfig base64 `-data` material, held IN MEMORY (never written
+  // to /tmp: the daemon re-resolves config every tick and leaked key files
+  // would accumulate unboundedly — ADVICE r1).
+  std::optional<std::string> ca_data;
+  std::optional<std::string> client_cert_data;
+  std::optional<std::string> client_key_data;
   bool skip_tls = false;
   std::string default_namespace = "default";
 

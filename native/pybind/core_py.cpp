@@ -266,6 +266,11 @@ PYBIND11_MODULE(_pruner_core, m) {
     d["token_file"] =
         cfg.token_file ? py::cast(*cfg.token_file) : py::object(py::none());
     d["ca_file"] = cfg.ca_file ? py::cast(*cfg.ca_file) : py::object(py::none());
+    d["ca_data"] = cfg.ca_data ? py::cast(*cfg.ca_data) : py::object(py::none());
+    d["client_cert_data"] =
+        cfg.client_cert_data ? py::cast(*cfg.client_cert_data) : py::object(py::none());
+    d["client_key_data"] =
+        cfg.client_key_data ? py::cast(*cfg.client_key_data) : py::object(py::none());
     d["skip_tls"] = cfg.skip_tls;
     d["default_namespace"] = cfg.default_namespace;
     return d;

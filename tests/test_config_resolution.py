@@ -132,18 +132,23 @@ users:
 
 def test_kubeconfig_token_auth(core, clean_env, tmp_path):
     import base64
+    import glob
 
     ca_pem = "-----BEGIN CERTIFICATE-----\nabc\n-----END CERTIFICATE-----\n"
     kc = tmp_path / "config"
     kc.write_text(KUBECONFIG_TMPL.format(
         ca_b64=base64.b64encode(ca_pem.encode()).decode()))
     clean_env.setenv("KUBECONFIG", str(kc))
+    tmp_before = set(glob.glob("/tmp/gpu-pruner-*"))
     cfg = core.resolve_kube_config()
     assert cfg["url"] == "https://api.prod.example:6443"
     assert cfg["token"] == "kubeconfig-token-123"
     assert cfg["default_namespace"] == "ml-team"
-    # -data CA decoded to a temp file
-    assert cfg["ca_file"] and open(cfg["ca_file"]).read() == ca_pem
+    # -data CA decoded IN MEMORY — never written to disk (key material must
+    # not accumulate in /tmp across the daemon's per-tick config re-resolves)
+    assert cfg["ca_file"] is None
+    assert cfg["ca_data"] == ca_pem
+    assert set(glob.glob("/tmp/gpu-pruner-*")) == tmp_before
 
 
 def test_kubeconfig_insecure_and_cert_files(core, clean_env, tmp_path):

@@ -114,8 +114,10 @@ def test_failure_breaker_exits_nonzero(pruner_bin, fake_api, fake_prom):
                    "--check-interval", "0", "--max-failures", "2", timeout=60)
     assert r.returncode != 0
     assert "Too many consecutive failures" in r.stderr
-    # 1 initial + 2 tolerated + 1 tripping = at least 3 queries attempted
-    assert len(fake_prom.queries) >= 3
+    # Reference parity (main.rs:310-321 compares the PRE-increment counter):
+    # with max 2 the breaker trips on the 4th consecutive failure — failures
+    # with pre-increment values 0, 1, 2 are tolerated, 3 > 2 exits.
+    assert len(fake_prom.queries) == 4
 
 
 def test_failure_then_recovery_does_not_trip(pruner_bin, fake_api, cluster, fake_prom):

@@ -85,7 +85,7 @@ users:
         monkeypatch.delenv(var, raising=False)
     monkeypatch.setenv("KUBECONFIG", str(kc))
     cfg = core.resolve_kube_config()
-    assert open(cfg["ca_file"], "rb").read() == pem
+    assert cfg["ca_data"].encode() == pem  # decoded in memory, not to a file
 
 
 hypothesis = pytest.importorskip("hypothesis")
