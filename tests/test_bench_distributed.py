@@ -50,7 +50,12 @@ def test_bench_single_rank_json_contract():
                 "dtype", "data", "config"):
         assert key in result, f"missing contract key {key}"
     assert result["data"] == "synthetic"
-    assert result["config"]["p50_scale_decision_latency_ms"] > 0
+    assert result["config"]["p50_tick_latency_ms"] > 0
+    assert result["config"]["per_pod_decision_latency_us"] > 0
+    # BASELINE config 5 shape by default: OTLP export on + an RTT-bound point
+    assert result["config"]["otlp"].startswith("enabled")
+    assert result["config"]["rtt_bound"]["apiserver_latency_us"] == 2000
+    assert result["config"]["rtt_bound"]["pods_per_sec"] > 0
 
 
 def test_bench_latency_injection_flag():
