@@ -89,10 +89,14 @@ def test_busy_probe_raises_utilization(sampler):
             busy = sampler.snapshot()[0]["busy_percent"]
             if busy >= 90.0:
                 break
+        assert busy >= 90.0, f"probe should saturate the GPU (busy={busy})"
+        # hold the load for a full 2 s sliding window before reading the ratio
+        for _ in range(22):
+            time.sleep(0.1)
+            sampler.poll_once()
+        ratio = sampler.snapshot()[0]["gr_engine_active"]
     finally:
         probe.stop()
-    assert busy >= 90.0, f"probe should saturate the GPU (busy={busy})"
-    ratio = sampler.snapshot()[0]["gr_engine_active"]
     assert ratio > 0.3, f"windowed ratio under load = {ratio}"
     # and utilization must fall back to zero afterwards
     busy = _settle_idle(sampler, seconds=10.0)
