@@ -2,6 +2,7 @@
 
 #include <unistd.h>
 
+#include <algorithm>
 #include <cstdio>
 #include <cstring>
 
@@ -13,12 +14,16 @@ namespace logx {
 namespace {
 
 struct State {
-  std::atomic<int> level{static_cast<int>(Level::Info)};
+  std::atomic<int> level{static_cast<int>(Level::Info)};  // min over all directives
+  std::atomic<int> default_level{static_cast<int>(Level::Info)};
   std::atomic<int> format{static_cast<int>(Format::Default)};
   std::atomic<bool> color{false};
   std::mutex write_mu;
   std::mutex counter_mu;
   std::map<std::string, int64_t> counters;
+  // Per-target directives (path → level), longest path first. Written only
+  // by init() (process startup, before worker threads); read lock-free.
+  std::vector<std::pair<std::string, int>> directives;
 };
 
 State& state() {
@@ -64,22 +69,56 @@ Level parse_level(const std::string& s, Level dflt) {
 void init(Format format, const char* env_filter) {
   auto& s = state();
   s.format.store(static_cast<int>(format));
-  Level lvl = Level::Info;
+  Level dflt = Level::Info;
+  std::vector<std::pair<std::string, int>> directives;
   const char* env = env_filter;
   if (!env || !*env) env = std::getenv("GPU_PRUNER_LOG");
   if (!env || !*env) env = std::getenv("RUST_LOG");  // drop-in with the reference's env knob
   if (env && *env) {
-    // accept "debug" or a comma list like "info,pruner=debug" (global part only)
-    std::string first = strutil::split(env, ',')[0];
-    if (first.find('=') == std::string::npos) lvl = parse_level(first, Level::Info);
+    // env_logger / tracing-subscriber EnvFilter directive list:
+    //   "info,pruner::engine=debug,hyper=error"
+    // bare level → default; path=level → that path and its ::-descendants.
+    for (const std::string& raw : strutil::split(env, ',')) {
+      std::string item = strutil::trim(raw);
+      if (item.empty()) continue;
+      size_t eq = item.find('=');
+      if (eq == std::string::npos) {
+        dflt = parse_level(item, dflt);
+      } else {
+        std::string path = strutil::trim(item.substr(0, eq));
+        Level lv = parse_level(item.substr(eq + 1), Level::Info);
+        if (!path.empty()) directives.emplace_back(path, static_cast<int>(lv));
+      }
+    }
   }
-  s.level.store(static_cast<int>(lvl));
+  // longest path first so the most specific directive wins
+  std::sort(directives.begin(), directives.end(),
+            [](const auto& a, const auto& b) { return a.first.size() > b.first.size(); });
+  int min_lvl = static_cast<int>(dflt);
+  for (const auto& [_, lv] : directives) min_lvl = std::min(min_lvl, lv);
+  s.directives = std::move(directives);
+  s.default_level.store(static_cast<int>(dflt));
+  s.level.store(min_lvl);
   s.color.store(isatty(2) != 0);
 }
 
 Level level() { return static_cast<Level>(state().level.load(std::memory_order_relaxed)); }
 
 bool enabled(Level lvl) { return static_cast<int>(lvl) >= state().level.load(std::memory_order_relaxed); }
+
+bool enabled_for(Level lvl, const std::string& target) {
+  auto& s = state();
+  for (const auto& [path, dir_lvl] : s.directives) {
+    // prefix match on module-path boundaries: "pruner" matches
+    // "pruner::engine" but not "prunerx"
+    if (target.size() >= path.size() && target.compare(0, path.size(), path) == 0 &&
+        (target.size() == path.size() ||
+         (target.size() >= path.size() + 2 && target[path.size()] == ':' &&
+          target[path.size() + 1] == ':')))
+      return static_cast<int>(lvl) >= dir_lvl;
+  }
+  return static_cast<int>(lvl) >= s.default_level.load(std::memory_order_relaxed);
+}
 
 void emit(Level lvl, const std::string& target, const std::string& msg) {
   emit_kv(lvl, target, msg, {});
@@ -88,7 +127,7 @@ void emit(Level lvl, const std::string& target, const std::string& msg) {
 void emit_kv(Level lvl, const std::string& target, const std::string& msg,
              const std::vector<std::pair<std::string, std::string>>& fields) {
   auto& s = state();
-  if (!enabled(lvl)) return;
+  if (!enabled_for(lvl, target)) return;
   Format fmt = static_cast<Format>(s.format.load(std::memory_order_relaxed));
   std::string line;
   std::string ts = strutil::rfc3339_micro_now();

@@ -26,10 +26,17 @@ struct Config {
 };
 
 // Initialize global logger. `env_filter` comes from GPU_PRUNER_LOG / RUST_LOG
-// ("trace".."error","off"); flag-level format selection mirrors --log-format.
+// and supports env_logger-style comma lists of per-target directives —
+// "info,pruner::engine=debug,hyper=error" — where a bare level sets the
+// default and `path=level` applies to `path` and its `::`-descendants, the
+// longest matching path winning (reference parity: tracing-subscriber
+// EnvFilter, main.rs:157-173). Flag-level format selection mirrors
+// --log-format.
 void init(Format format, const char* env_filter = nullptr);
 Level level();
-bool enabled(Level lvl);
+bool enabled(Level lvl);  // fast path: true if ANY target could log at lvl
+// Per-target decision honoring directives.
+bool enabled_for(Level lvl, const std::string& target);
 
 // Core emit. `target` is the module path shown in logs (e.g. "pruner::engine").
 void emit(Level lvl, const std::string& target, const std::string& msg);
@@ -41,9 +48,10 @@ void emit_kv(Level lvl, const std::string& target, const std::string& msg,
 // The message expression is only evaluated when the level is enabled — hot
 // loops log per pod, and the string concatenation would otherwise dominate
 // filtered-out levels.
-#define LOGX_AT(lvl, target, msg)                                   \
-  do {                                                              \
-    if (::logx::enabled(lvl)) ::logx::emit(lvl, target, msg);       \
+#define LOGX_AT(lvl, target, msg)                                          \
+  do {                                                                     \
+    if (::logx::enabled(lvl) && ::logx::enabled_for(lvl, target))          \
+      ::logx::emit(lvl, target, msg);                                      \
   } while (0)
 #define LOGT(target, msg) LOGX_AT(::logx::Level::Trace, target, msg)
 #define LOGD(target, msg) LOGX_AT(::logx::Level::Debug, target, msg)

@@ -106,6 +106,40 @@ def test_log_level_filter_env(pruner_bin, idle_cluster, fake_prom):
     assert not any(" INFO " in l for l in r2.stderr.splitlines())
 
 
+def test_log_per_target_directives(pruner_bin, idle_cluster, fake_prom):
+    """env_logger-style per-target filtering (reference EnvFilter,
+    main.rs:157-173): 'error,pruner::engine=info' silences every target
+    except the engine."""
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom,
+                   env_extra={"GPU_PRUNER_LOG": "error,pruner::engine=info"})
+    assert r.returncode == 0, r.stderr
+    info = [l for l in r.stderr.splitlines() if " INFO " in l]
+    assert info, "engine INFO lines must pass the directive"
+    assert all("pruner::engine" in l for l in info), info
+    # no daemon/prom INFO chatter ("Query succeeded" is pruner::daemon INFO)
+    assert "Query succeeded" not in r.stderr
+
+
+def test_log_directive_prefix_matches_module_boundary(pruner_bin, idle_cluster,
+                                                      fake_prom):
+    """'pruner=debug' covers pruner::engine / pruner::daemon descendants."""
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom,
+                   env_extra={"GPU_PRUNER_LOG": "off,pruner=info"})
+    assert r.returncode == 0, r.stderr
+    assert "Query succeeded" in r.stderr  # pruner::daemon INFO
+    assert "pruner::engine" in r.stderr
+
+
+def test_log_most_specific_directive_wins(pruner_bin, idle_cluster, fake_prom):
+    r = run_daemon(pruner_bin, idle_cluster, fake_prom,
+                   env_extra={"GPU_PRUNER_LOG": "pruner=info,pruner::engine=error"})
+    assert r.returncode == 0, r.stderr
+    assert "Query succeeded" in r.stderr  # pruner::daemon stays at info
+    engine_info = [l for l in r.stderr.splitlines()
+                   if " INFO " in l and "pruner::engine" in l]
+    assert not engine_info, engine_info
+
+
 def test_log_format_pretty(pruner_bin, idle_cluster, fake_prom):
     r = run_daemon(pruner_bin, idle_cluster, fake_prom, "--log-format", "pretty")
     assert r.returncode == 0
