@@ -64,6 +64,11 @@ class FakeApiServer:
         self.requests: list[tuple[str, str]] = []  # (method, path)
         self.token = token
         self.latency_s = latency_s  # simulated apiserver RTT for benchmarks
+        # throttle injection: respond 429 (+ Retry-After) to the next N
+        # requests — exercises the client's bounded-retry path
+        self.throttle_next = 0
+        self.retry_after_s = 0
+        self.throttled = 0  # how many 429s were actually served
 
         fixture = self
 
@@ -74,13 +79,27 @@ class FakeApiServer:
             def log_message(self, *args):
                 pass
 
-            def _send(self, status: int, obj):
+            def _send(self, status: int, obj, headers=None):
                 body = json.dumps(obj).encode()
                 self.send_response(status)
                 self.send_header("Content-Type", "application/json")
                 self.send_header("Content-Length", str(len(body)))
+                for k, v in (headers or {}).items():
+                    self.send_header(k, v)
                 self.end_headers()
                 self.wfile.write(body)
+
+            def _maybe_throttle(self) -> bool:
+                with fixture._lock:
+                    if fixture.throttle_next <= 0:
+                        return False
+                    fixture.throttle_next -= 1
+                    fixture.throttled += 1
+                    ra = fixture.retry_after_s
+                self._send(429, {"kind": "Status", "code": 429,
+                                 "reason": "TooManyRequests"},
+                           headers={"Retry-After": str(ra)})
+                return True
 
             def _auth_ok(self) -> bool:
                 if fixture.token is None:
@@ -103,6 +122,8 @@ class FakeApiServer:
                     fixture.requests.append(("GET", self.path))
                 if fixture.latency_s:
                     time.sleep(fixture.latency_s)
+                if self._maybe_throttle():
+                    return
                 if not self._auth_ok():
                     return self._send(401, {"kind": "Status", "code": 401})
                 # cluster-scope pod list (used by the exporter's attribution
@@ -140,6 +161,8 @@ class FakeApiServer:
                     fixture.requests.append(("PATCH", self.path))
                 if fixture.latency_s:
                     time.sleep(fixture.latency_s)
+                if self._maybe_throttle():
+                    return
                 if not self._auth_ok():
                     return self._send(401, {"kind": "Status", "code": 401})
                 length = int(self.headers.get("Content-Length", "0"))
@@ -173,6 +196,8 @@ class FakeApiServer:
                     fixture.requests.append(("POST", self.path))
                 if fixture.latency_s:
                     time.sleep(fixture.latency_s)
+                if self._maybe_throttle():
+                    return
                 if not self._auth_ok():
                     return self._send(401, {"kind": "Status", "code": 401})
                 length = int(self.headers.get("Content-Length", "0"))
