@@ -1,8 +1,11 @@
 #include "k8s.hpp"
 
 #include <chrono>
+#include <cstdio>
 #include <cstdlib>
+#include <ctime>
 #include <fstream>
+#include <map>
 #include <thread>
 
 #include "../common/log.hpp"
@@ -75,6 +78,17 @@ std::optional<KubeConfig> load_kubeconfig(const std::string& path) {
       cfg.ca_data = pem;
   }
   if (user.is_object()) {
+    if (user.get("exec").is_object()) {
+      const jsn::Value& ex = user.get("exec");
+      ExecConfig ec;
+      ec.command = ex.get("command").as_string();
+      if (ex.get("args").is_array())
+        for (const auto& a : ex.get("args").arr()) ec.args.push_back(a.as_string());
+      if (ex.get("env").is_array())
+        for (const auto& e : ex.get("env").arr())
+          ec.env.emplace_back(e.get("name").as_string(), e.get("value").as_string());
+      if (!ec.command.empty()) cfg.exec = std::move(ec);
+    }
     if (user.get("token").is_string()) cfg.token = user.get("token").as_string();
     if (user.get("tokenFile").is_string()) cfg.token_file = user.get("tokenFile").as_string();
     if (user.get("client-certificate").is_string())
@@ -97,7 +111,98 @@ std::optional<KubeConfig> load_kubeconfig(const std::string& path) {
   return cfg;
 }
 
+// ---- exec credential plugin (client.authentication.k8s.io) ------------------
+//
+// Runs the kubeconfig's `user.exec` command and parses the ExecCredential
+// JSON from stdout:
+//   {"kind":"ExecCredential","status":{"token":"...",
+//    "expirationTimestamp":"2026-01-01T00:00:00Z"}}
+// Tokens are cached per command line until expiry (minus a 60 s skew margin)
+// so the daemon's per-tick client rebuild does not fork a plugin every tick.
+
+struct ExecCredCache {
+  std::mutex mu;
+  std::map<std::string, std::pair<std::string, double>> entries;  // key → (token, expiry)
+};
+
+ExecCredCache& exec_cache() {
+  static ExecCredCache c;
+  return c;
+}
+
+std::string shell_quote(const std::string& s) {
+  std::string out = "'";
+  for (char c : s) {
+    if (c == '\'') out += "'\\''";
+    else out += c;
+  }
+  out += "'";
+  return out;
+}
+
+std::optional<std::string> run_exec_plugin(const ExecConfig& ec) {
+  std::string cmdline;
+  for (const auto& [k, v] : ec.env) cmdline += k + "=" + shell_quote(v) + " ";
+  cmdline += shell_quote(ec.command);
+  for (const auto& a : ec.args) cmdline += " " + shell_quote(a);
+
+  std::string key = cmdline;
+  double now = static_cast<double>(::time(nullptr));
+  {
+    auto& cache = exec_cache();
+    std::lock_guard<std::mutex> lock(cache.mu);
+    auto it = cache.entries.find(key);
+    if (it != cache.entries.end() && (it->second.second == 0.0 || now < it->second.second))
+      return it->second.first;
+  }
+
+  std::string out;
+  cmdline += " 2>/dev/null";
+  FILE* p = ::popen(cmdline.c_str(), "r");
+  if (!p) {
+    LOGW("pruner::k8s", "failed to spawn exec credential plugin: " + ec.command);
+    return std::nullopt;
+  }
+  char buf[4096];
+  size_t n;
+  while ((n = ::fread(buf, 1, sizeof buf, p)) > 0) out.append(buf, n);
+  int rc = ::pclose(p);
+  if (rc != 0) {
+    LOGW("pruner::k8s", "exec credential plugin " + ec.command + " exited " +
+                            std::to_string(rc));
+    return std::nullopt;
+  }
+  try {
+    jsn::Value cred = jsn::parse(out);
+    const jsn::Value& status = cred.get("status");
+    std::string token = status.get("token").as_string();
+    if (token.empty()) {
+      LOGW("pruner::k8s", "exec credential plugin returned no status.token");
+      return std::nullopt;
+    }
+    double expiry = 0.0;  // 0 = no expiry given; cache until process exit
+    if (status.get("expirationTimestamp").is_string()) {
+      double ts;
+      if (strutil::parse_rfc3339(status.get("expirationTimestamp").as_string(), &ts))
+        expiry = ts - 60.0;  // refresh 60 s before the deadline
+    }
+    auto& cache = exec_cache();
+    std::lock_guard<std::mutex> lock(cache.mu);
+    cache.entries[key] = {token, expiry};
+    return token;
+  } catch (const std::exception& e) {
+    LOGW("pruner::k8s", std::string("bad ExecCredential JSON from plugin: ") + e.what());
+    return std::nullopt;
+  }
+}
+
 }  // namespace
+
+void exec_cred_cache_clear_for_test() {
+  auto& cache = exec_cache();
+  std::lock_guard<std::mutex> lock(cache.mu);
+  cache.entries.clear();
+}
 
 KubeConfig KubeConfig::resolve() {
   KubeConfig cfg;
@@ -169,6 +274,9 @@ std::string KubeClient::bearer() const {
   if (cfg_.token) return *cfg_.token;
   if (cfg_.token_file) {
     if (auto t = read_file(*cfg_.token_file)) return strutil::trim(*t);
+  }
+  if (cfg_.exec) {
+    if (auto t = run_exec_plugin(*cfg_.exec)) return *t;
   }
   return "";
 }

@@ -17,12 +17,24 @@
 #include <memory>
 #include <optional>
 #include <string>
+#include <vector>
 
 #include "../common/http.hpp"
 #include "../common/json.hpp"
 #include "resources.hpp"
 
 namespace pruner {
+
+// kubeconfig `user.exec` credential plugin (client.authentication.k8s.io):
+// managed clusters (EKS aws-iam-authenticator, GKE gke-gcloud-auth-plugin,
+// OpenShift oc) mint short-lived tokens through an external command whose
+// stdout is an ExecCredential JSON. Tokens are cached until their
+// expirationTimestamp.
+struct ExecConfig {
+  std::string command;
+  std::vector<std::string> args;
+  std::vector<std::pair<std::string, std::string>> env;
+};
 
 struct KubeConfig {
   std::string url;                 // https://host:port
@@ -38,6 +50,7 @@ struct KubeConfig {
   std::optional<std::string> ca_data;
   std::optional<std::string> client_cert_data;
   std::optional<std::string> client_key_data;
+  std::optional<ExecConfig> exec;  // credential plugin (run lazily, cached)
   bool skip_tls = false;
   std::string default_namespace = "default";
 
@@ -45,6 +58,9 @@ struct KubeConfig {
   // override nor in-cluster config is present.
   static KubeConfig resolve();
 };
+
+// Drop cached exec-plugin tokens (tests exercise expiry/refresh).
+void exec_cred_cache_clear_for_test();
 
 class KubeError : public std::runtime_error {
 public:

@@ -273,6 +273,8 @@ PYBIND11_MODULE(_pruner_core, m) {
         cfg.client_key_data ? py::cast(*cfg.client_key_data) : py::object(py::none());
     d["skip_tls"] = cfg.skip_tls;
     d["default_namespace"] = cfg.default_namespace;
+    d["exec_command"] =
+        cfg.exec ? py::cast(cfg.exec->command) : py::object(py::none());
     return d;
   });
 
