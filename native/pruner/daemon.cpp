@@ -83,25 +83,19 @@ int run_daemon(const Config& cfg) {
   std::mutex consumer_kube_mu;
   auto consume = [&] {
     while (auto sk = queue.pop()) {
-      if (!(enabled & kind_flag(sk->kind))) {
-        LOGI(TARGET, "Skipping resource type " + sk->kind_str() + " because it is not enabled");
-        continue;
-      }
       try {
         {
           std::lock_guard<std::mutex> lock(consumer_kube_mu);
           if (!consumer_kube)
             consumer_kube = std::make_shared<KubeClient>(KubeConfig::resolve());
         }
-        scale(*consumer_kube, *sk);
       } catch (const std::exception& e) {
         logx::counter_add("monotonic_counter.scale_failures", 1);
-        LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
+        LOGE(TARGET, std::string("Failed to build Kubernetes client: ") + e.what());
         continue;
       }
-      logx::counter_add("monotonic_counter.scale_successes", 1);
-      LOGI(TARGET, "Scaled Resource: [" + sk->kind_str() + "] - " +
-                       sk->ns().value_or("default") + ":" + sk->name());
+      // shared actuation path (enabled-mask check + scale + counters + log)
+      scale_one(*consumer_kube, *sk, enabled);
     }
   };
   int n_consumers = std::max(1, std::min(cfg.max_concurrency, 8));

@@ -154,25 +154,31 @@ void scale(KubeClient& kube, const ScaleKind& sk) {
   }
 }
 
+bool scale_one(KubeClient& kube, const ScaleKind& sk, uint8_t enabled_mask) {
+  if (!(enabled_mask & kind_flag(sk.kind))) {
+    LOGI(TARGET, "Skipping resource type " + sk.kind_str() + " because it is not enabled");
+    return false;
+  }
+  try {
+    scale(kube, sk);
+  } catch (const std::exception& e) {
+    logx::counter_add("monotonic_counter.scale_failures", 1);
+    LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
+    return false;
+  }
+  logx::counter_add("monotonic_counter.scale_successes", 1);
+  LOGI(TARGET, "Scaled Resource: [" + sk.kind_str() + "] - " +
+                   sk.ns().value_or("default") + ":" + sk.name());
+  return true;
+}
+
 size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
                  uint8_t enabled_mask, int concurrency) {
   std::atomic<size_t> scaled{0};
   qx::ThreadPool::global().parallel_for(
       roots.size(), concurrency, [&](size_t i) {
-        const ScaleKind& sk = roots[i];
-        if (!(enabled_mask & kind_flag(sk.kind))) {
-          LOGI(TARGET,
-               "Skipping resource type " + sk.kind_str() + " because it is not enabled");
-          return;
-        }
-        try {
-          scale(kube, sk);
+        if (scale_one(kube, roots[i], enabled_mask))
           scaled.fetch_add(1, std::memory_order_relaxed);
-          logx::counter_add("monotonic_counter.scale_successes", 1);
-        } catch (const std::exception& e) {
-          logx::counter_add("monotonic_counter.scale_failures", 1);
-          LOGE(TARGET, std::string("Failed to scale resource! ") + e.what());
-        }
       });
   return scaled.load();
 }

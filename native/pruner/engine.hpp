@@ -47,6 +47,11 @@ void scale(KubeClient& kube, const ScaleKind& sk);
 // successes, logs failures. The reference drains its scale channel with a
 // single serial consumer (main.rs:332-367); with per-root RTTs this is the
 // second fan-out that matters at 1000-pod scale.
+// Single actuation step shared by the daemon's consumer pool and scale_all
+// (one path for enabled-mask check + scale + counters).
+// Returns true when the resource was actually scaled.
+bool scale_one(KubeClient& kube, const ScaleKind& sk, uint8_t enabled_mask);
+
 size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
                  uint8_t enabled_mask, int concurrency);
 
