@@ -10,8 +10,8 @@ CPU, plus a synthetic 1000-pod cluster generator for the benchmark configs.
 from .fake_prom import FakePrometheus
 from .fake_apiserver import FakeApiServer
 from .synth import build_synthetic_cluster
-from .fake_otlp import FakeOtlpCollector
+from .fake_otlp import FakeOtlpCollector, FakeOtlpGrpcCollector
 from .miniprom import MiniProm
 
 __all__ = ["FakePrometheus", "FakeApiServer", "build_synthetic_cluster",
-           "FakeOtlpCollector", "MiniProm"]
+           "FakeOtlpCollector", "FakeOtlpGrpcCollector", "MiniProm"]

@@ -104,3 +104,67 @@ class FakeOtlpCollector:
                             if dps:
                                 out[m["name"]] = int(dps[-1].get("asInt", 0))
             return out
+
+    def spans(self) -> list[dict]:
+        """All exported span dicts (JSON transport)."""
+        with self._lock:
+            out = []
+            for payload in self.traces:
+                for rs in payload.get("resourceSpans", []):
+                    for ss in rs.get("scopeSpans", []):
+                        out.extend(ss.get("spans", []))
+            return out
+
+
+class FakeOtlpGrpcCollector:
+    """OTLP/gRPC collector double (the reference's tonic transport): a real
+    grpcio server accepting TraceService/MetricsService Export as raw bytes —
+    an independent HTTP/2 + gRPC implementation validating the hand-rolled
+    h2c client (native/common/grpc_client.cpp)."""
+
+    def __init__(self, host: str = "127.0.0.1", port: int = 0):
+        import grpc
+        from concurrent import futures
+
+        self._lock = threading.Lock()
+        self.traces_pb: list[bytes] = []
+        self.metrics_pb: list[bytes] = []
+        fixture = self
+
+        class Handler(grpc.GenericRpcHandler):
+            def service(self, handler_call_details):
+                method = handler_call_details.method
+
+                def unary_unary(request, context):
+                    with fixture._lock:
+                        if "TraceService" in method:
+                            fixture.traces_pb.append(request)
+                        elif "MetricsService" in method:
+                            fixture.metrics_pb.append(request)
+                    return b""  # empty Export*ServiceResponse
+
+                return grpc.unary_unary_rpc_method_handler(
+                    unary_unary, request_deserializer=None,
+                    response_serializer=None)
+
+        self._server = grpc.server(futures.ThreadPoolExecutor(max_workers=4),
+                                   handlers=(Handler(),))
+        self._port = self._server.add_insecure_port(f"{host}:{port}")
+        self._host = host
+
+    def start(self):
+        self._server.start()
+        return self
+
+    def stop(self):
+        self._server.stop(grace=None)
+
+    @property
+    def url(self) -> str:
+        return f"http://{self._host}:{self._port}"
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()

@@ -8,6 +8,7 @@
 #include <chrono>
 #include <cstring>
 
+#include "../common/grpc_client.hpp"
 #include "../common/log.hpp"
 
 namespace exporter {
@@ -15,135 +16,6 @@ namespace exporter {
 namespace {
 
 constexpr const char* TARGET = "exporter::podresources";
-
-// ---------------- raw socket with deadline ----------------
-
-class Sock {
-public:
-  Sock(const std::string& path, int timeout_ms)
-      : deadline_(std::chrono::steady_clock::now() + std::chrono::milliseconds(timeout_ms)) {
-    fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
-    if (fd_ < 0) throw PodResourcesError("socket(AF_UNIX) failed");
-    struct sockaddr_un addr {};
-    addr.sun_family = AF_UNIX;
-    if (path.size() >= sizeof(addr.sun_path)) {
-      ::close(fd_);
-      throw PodResourcesError("socket path too long: " + path);
-    }
-    std::strncpy(addr.sun_path, path.c_str(), sizeof(addr.sun_path) - 1);
-    if (::connect(fd_, reinterpret_cast<struct sockaddr*>(&addr), sizeof addr) < 0) {
-      ::close(fd_);
-      throw PodResourcesError("connect to " + path + " failed: " + std::strerror(errno));
-    }
-  }
-  ~Sock() {
-    if (fd_ >= 0) ::close(fd_);
-  }
-
-  void write_all(const void* data, size_t n) {
-    const char* p = static_cast<const char*>(data);
-    size_t off = 0;
-    while (off < n) {
-      ssize_t w = ::send(fd_, p + off, n - off, MSG_NOSIGNAL);
-      if (w < 0) {
-        if (errno == EINTR) continue;
-        if (errno == EAGAIN || errno == EWOULDBLOCK) {
-          wait_io(false);
-          continue;
-        }
-        throw PodResourcesError(std::string("write failed: ") + std::strerror(errno));
-      }
-      off += static_cast<size_t>(w);
-    }
-  }
-
-  // Exactly n bytes or throw.
-  void read_exact(void* buf, size_t n) {
-    char* p = static_cast<char*>(buf);
-    size_t off = 0;
-    while (off < n) {
-      ssize_t r = ::recv(fd_, p + off, n - off, MSG_DONTWAIT);
-      if (r > 0) {
-        off += static_cast<size_t>(r);
-        continue;
-      }
-      if (r == 0) throw PodResourcesError("connection closed mid-frame");
-      if (errno == EINTR) continue;
-      if (errno == EAGAIN || errno == EWOULDBLOCK) {
-        wait_io(true);
-        continue;
-      }
-      throw PodResourcesError(std::string("read failed: ") + std::strerror(errno));
-    }
-  }
-
-private:
-  void wait_io(bool want_read) {
-    auto now = std::chrono::steady_clock::now();
-    if (now >= deadline_) throw PodResourcesError("PodResources call timed out");
-    int ms = static_cast<int>(
-        std::chrono::duration_cast<std::chrono::milliseconds>(deadline_ - now).count());
-    struct pollfd pfd {fd_, static_cast<short>(want_read ? POLLIN : POLLOUT), 0};
-    int rc = ::poll(&pfd, 1, std::max(ms, 1));
-    if (rc == 0) throw PodResourcesError("PodResources call timed out");
-    if (rc < 0 && errno != EINTR)
-      throw PodResourcesError(std::string("poll failed: ") + std::strerror(errno));
-  }
-
-  int fd_ = -1;
-  std::chrono::steady_clock::time_point deadline_;
-};
-
-// ---------------- HTTP/2 framing ----------------
-
-enum FrameType : uint8_t {
-  F_DATA = 0x0,
-  F_HEADERS = 0x1,
-  F_RST_STREAM = 0x3,
-  F_SETTINGS = 0x4,
-  F_PING = 0x6,
-  F_GOAWAY = 0x7,
-  F_WINDOW_UPDATE = 0x8,
-  F_CONTINUATION = 0x9,
-};
-
-constexpr uint8_t FLAG_END_STREAM = 0x1;
-constexpr uint8_t FLAG_ACK = 0x1;
-
-void put_frame_header(std::string& out, size_t len, uint8_t type, uint8_t flags,
-                      uint32_t stream) {
-  out += static_cast<char>((len >> 16) & 0xFF);
-  out += static_cast<char>((len >> 8) & 0xFF);
-  out += static_cast<char>(len & 0xFF);
-  out += static_cast<char>(type);
-  out += static_cast<char>(flags);
-  out += static_cast<char>((stream >> 24) & 0x7F);
-  out += static_cast<char>((stream >> 16) & 0xFF);
-  out += static_cast<char>((stream >> 8) & 0xFF);
-  out += static_cast<char>(stream & 0xFF);
-}
-
-// HPACK emitters (request side only): static-table indexed fields and
-// literal-without-indexing fields — no dynamic table, no Huffman.
-void hpack_indexed(std::string& out, uint8_t index) {
-  out += static_cast<char>(0x80 | index);
-}
-void hpack_str(std::string& out, const std::string& s) {
-  // 7-bit length prefix, no Huffman; all our strings are < 127 bytes
-  out += static_cast<char>(s.size() & 0x7F);
-  out += s;
-}
-void hpack_literal_indexed_name(std::string& out, uint8_t name_index,
-                                const std::string& value) {
-  out += static_cast<char>(name_index & 0x0F);  // 0000xxxx: without indexing
-  hpack_str(out, value);
-}
-void hpack_literal_new_name(std::string& out, const std::string& name,
-                            const std::string& value) {
-  out += static_cast<char>(0x00);
-  hpack_str(out, name);
-  hpack_str(out, value);
-}
 
 // ---------------- protobuf wire decoding ----------------
 
@@ -259,111 +131,18 @@ std::vector<PodResourcesEntry> decode_list_response(const std::string& payload) 
 
 std::vector<PodResourcesEntry> list_pod_resources(const std::string& socket_path,
                                                   int timeout_ms) {
-  Sock sock(socket_path, timeout_ms);
-
-  // ---- connection preface + SETTINGS + generous connection window ----
-  std::string out("PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n");
-  // SETTINGS: INITIAL_WINDOW_SIZE (0x4) = 16 MiB so large responses flow
-  // without per-stream WINDOW_UPDATE bookkeeping
-  put_frame_header(out, 6, F_SETTINGS, 0, 0);
-  out += static_cast<char>(0x00);
-  out += static_cast<char>(0x04);
-  uint32_t win = 1u << 24;
-  out += static_cast<char>((win >> 24) & 0xFF);
-  out += static_cast<char>((win >> 16) & 0xFF);
-  out += static_cast<char>((win >> 8) & 0xFF);
-  out += static_cast<char>(win & 0xFF);
-  // connection-level WINDOW_UPDATE: +16 MiB
-  put_frame_header(out, 4, F_WINDOW_UPDATE, 0, 0);
-  out += static_cast<char>((win >> 24) & 0x7F);
-  out += static_cast<char>((win >> 16) & 0xFF);
-  out += static_cast<char>((win >> 8) & 0xFF);
-  out += static_cast<char>(win & 0xFF);
-
-  // ---- HEADERS (stream 1) ----
-  std::string hdrs;
-  hpack_indexed(hdrs, 3);                               // :method: POST
-  hpack_indexed(hdrs, 6);                               // :scheme: http
-  hpack_literal_indexed_name(hdrs, 4, "/v1.PodResourcesLister/List");  // :path
-  hpack_literal_indexed_name(hdrs, 1, "localhost");     // :authority
-  hpack_literal_new_name(hdrs, "content-type", "application/grpc");
-  hpack_literal_new_name(hdrs, "te", "trailers");
-  put_frame_header(out, hdrs.size(), F_HEADERS, 0x4 /*END_HEADERS*/, 1);
-  out += hdrs;
-
-  // ---- DATA: gRPC frame carrying the empty ListPodResourcesRequest ----
-  const char grpc_empty[5] = {0, 0, 0, 0, 0};
-  put_frame_header(out, 5, F_DATA, FLAG_END_STREAM, 1);
-  out.append(grpc_empty, 5);
-
-  sock.write_all(out.data(), out.size());
-
-  // ---- read frames until END_STREAM on stream 1 ----
-  std::string grpc_payload;
-  bool stream_done = false;
-  while (!stream_done) {
-    uint8_t fh[9];
-    sock.read_exact(fh, 9);
-    size_t len = (static_cast<size_t>(fh[0]) << 16) | (static_cast<size_t>(fh[1]) << 8) | fh[2];
-    uint8_t type = fh[3], flags = fh[4];
-    uint32_t stream = (static_cast<uint32_t>(fh[5] & 0x7F) << 24) |
-                      (static_cast<uint32_t>(fh[6]) << 16) |
-                      (static_cast<uint32_t>(fh[7]) << 8) | fh[8];
-    std::string payload(len, '\0');
-    if (len) sock.read_exact(payload.data(), len);
-
-    switch (type) {
-      case F_SETTINGS:
-        if (!(flags & FLAG_ACK)) {  // ack the server's settings
-          std::string ack;
-          put_frame_header(ack, 0, F_SETTINGS, FLAG_ACK, 0);
-          sock.write_all(ack.data(), ack.size());
-        }
-        break;
-      case F_PING:
-        if (!(flags & FLAG_ACK)) {
-          std::string pong;
-          put_frame_header(pong, 8, F_PING, FLAG_ACK, 0);
-          pong += payload;
-          sock.write_all(pong.data(), pong.size());
-        }
-        break;
-      case F_DATA:
-        if (stream == 1) grpc_payload += payload;
-        if (stream == 1 && (flags & FLAG_END_STREAM)) stream_done = true;
-        break;
-      case F_HEADERS:  // response headers / trailers — content not needed
-        if (stream == 1 && (flags & FLAG_END_STREAM)) stream_done = true;
-        break;
-      case F_RST_STREAM:
-        if (stream == 1) throw PodResourcesError("stream reset by kubelet");
-        break;
-      case F_GOAWAY:
-        if (!stream_done && grpc_payload.empty())
-          throw PodResourcesError("connection closed by kubelet (GOAWAY)");
-        stream_done = true;
-        break;
-      default:
-        break;  // WINDOW_UPDATE / CONTINUATION-free responses / unknown
-    }
+  // Unary v1.PodResourcesLister/List over the kubelet's unix socket, via the
+  // shared h2c gRPC client (native/common/grpc_client.cpp). The request is
+  // the empty ListPodResourcesRequest.
+  grpcx::Target t;
+  t.unix_path = socket_path;
+  std::string payload;
+  try {
+    payload = grpcx::unary_call(t, "/v1.PodResourcesLister/List", "", timeout_ms);
+  } catch (const grpcx::GrpcError& e) {
+    throw PodResourcesError(e.what());
   }
-
-  // ---- unwrap gRPC length-prefixed message(s) ----
-  std::vector<PodResourcesEntry> entries;
-  size_t pos = 0;
-  while (pos + 5 <= grpc_payload.size()) {
-    uint8_t compressed = static_cast<uint8_t>(grpc_payload[pos]);
-    uint32_t mlen = (static_cast<uint32_t>(static_cast<uint8_t>(grpc_payload[pos + 1])) << 24) |
-                    (static_cast<uint32_t>(static_cast<uint8_t>(grpc_payload[pos + 2])) << 16) |
-                    (static_cast<uint32_t>(static_cast<uint8_t>(grpc_payload[pos + 3])) << 8) |
-                    static_cast<uint8_t>(grpc_payload[pos + 4]);
-    if (compressed) throw PodResourcesError("compressed gRPC response unsupported");
-    if (pos + 5 + mlen > grpc_payload.size())
-      throw PodResourcesError("truncated gRPC message");
-    auto part = decode_list_response(grpc_payload.substr(pos + 5, mlen));
-    entries.insert(entries.end(), part.begin(), part.end());
-    pos += 5 + mlen;
-  }
+  std::vector<PodResourcesEntry> entries = decode_list_response(payload);
   LOGD(TARGET, "PodResources List: " + std::to_string(entries.size()) + " container entries");
   return entries;
 }

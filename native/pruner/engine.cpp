@@ -175,8 +175,12 @@ bool scale_one(KubeClient& kube, const ScaleKind& sk, uint8_t enabled_mask) {
 size_t scale_all(KubeClient& kube, const std::vector<ScaleKind>& roots,
                  uint8_t enabled_mask, int concurrency) {
   std::atomic<size_t> scaled{0};
+  // propagate the caller's span (run_query_and_scale) onto the pool threads
+  // so scale → scale_to_zero trees nest under it
+  otlp::SpanContext ctx = otlp::current_context();
   qx::ThreadPool::global().parallel_for(
       roots.size(), concurrency, [&](size_t i) {
+        otlp::ContextGuard cg(ctx);
         if (scale_one(kube, roots[i], enabled_mask))
           scaled.fetch_add(1, std::memory_order_relaxed);
       });
@@ -223,8 +227,10 @@ std::vector<ScaleKind> evaluate_candidates(KubeClient& kube, const jsn::Value& r
   // Evaluate pods concurrently: each needs 1-3 apiserver round-trips (pod GET
   // + owner walk). Worker count is the --max-concurrency knob.
   std::vector<std::optional<ScaleKind>> results(unique_pods.size());
+  otlp::SpanContext span_ctx = otlp::current_context();
   qx::ThreadPool::global().parallel_for(
       unique_pods.size(), cfg.max_concurrency, [&](size_t i) {
+      otlp::ContextGuard cg(span_ctx);
       const PodMetricData& pmd = unique_pods[i];
       try {
         auto pod = objs.get_pod(pmd.ns, pmd.name);

@@ -8,6 +8,7 @@
 #include <thread>
 #include <vector>
 
+#include "../common/grpc_client.hpp"
 #include "../common/http.hpp"
 #include "../common/pb.hpp"
 #include "../common/json.hpp"
@@ -20,6 +21,9 @@ namespace {
 
 struct FinishedSpan {
   std::string name;
+  std::string trace_id;   // 32 hex
+  std::string span_id;    // 16 hex
+  std::string parent_id;  // 16 hex or empty (root)
   uint64_t start_ns;
   uint64_t end_ns;
 };
@@ -27,23 +31,36 @@ struct FinishedSpan {
 constexpr size_t kMaxBufferedSpans = 50000;  // drop-oldest beyond this
 constexpr size_t kMaxSpansPerPost = 5000;    // bound per-request payload
 
+enum class Transport { Grpc, HttpProtobuf, HttpJson };
+
 struct State {
   std::atomic<bool> enabled{false};
   std::atomic<uint64_t> dropped_spans{0};
   std::atomic<bool> running{false};
   std::atomic<uint64_t> delivered{0};
   std::string endpoint;  // base, no trailing slash
+  std::string grpc_host;
+  uint16_t grpc_port = 4317;
   std::string service_name;
-  std::string trace_id;  // one id per process run; spans are flat (no parent tracking)
   std::mutex mu;
   std::vector<FinishedSpan> spans;
   std::thread exporter;
   std::condition_variable cv;
   bool stop = false;
   int interval_ms = 5000;
-  bool use_protobuf = false;  // OTEL_EXPORTER_OTLP_PROTOCOL=http/protobuf
+  Transport transport = Transport::HttpJson;
   std::unique_ptr<http::Client> client;  // persistent export connection pool
 };
+
+// Per-thread span lineage: innermost live SpanGuard = back of the stack.
+// ContextGuard seeds `installed_parent` so pool workers attach to the
+// caller's span instead of starting disconnected traces.
+struct ThreadCtx {
+  std::string trace_id;
+  std::vector<std::string> stack;
+  std::string installed_parent;
+};
+thread_local ThreadCtx t_ctx;
 
 State& state() {
   static State s;
@@ -100,13 +117,15 @@ std::string encode_spans_pb(const State& s, const std::vector<FinishedSpan>& bat
   std::string spans;
   for (const auto& fs : batch) {
     std::string sp;  // Span
-    pb::put_bytes(sp, 1, hex_to_bytes(s.trace_id));  // trace_id (16 bytes)
-    pb::put_bytes(sp, 2, hex_to_bytes(rand_hex(8))); // span_id (8 bytes)
-    pb::put_bytes(sp, 5, fs.name);                   // name
-    pb::put_varint(sp, 6, 1);                        // kind = INTERNAL
+    pb::put_bytes(sp, 1, hex_to_bytes(fs.trace_id));  // trace_id (16 bytes)
+    pb::put_bytes(sp, 2, hex_to_bytes(fs.span_id));   // span_id (8 bytes)
+    if (!fs.parent_id.empty())
+      pb::put_bytes(sp, 4, hex_to_bytes(fs.parent_id));  // parent_span_id
+    pb::put_bytes(sp, 5, fs.name);                    // name
+    pb::put_varint(sp, 6, 1);                         // kind = INTERNAL
     pb::put_fixed64(sp, 7, fs.start_ns);
     pb::put_fixed64(sp, 8, fs.end_ns);
-    pb::put_bytes(spans, 2, sp);                     // ScopeSpans.spans
+    pb::put_bytes(spans, 2, sp);                      // ScopeSpans.spans
   }
   std::string scope_spans;
   pb::put_bytes(scope_spans, 1, pb_scope());
@@ -191,6 +210,17 @@ void post_json(const std::string& url, const jsn::Value& body) {
   post_payload(url, body.dump(), "application/json");
 }
 
+// Unary OTLP/gRPC Export (the reference's tonic transport, main.rs:206-221):
+// ExportTrace/MetricsServiceRequest protobuf over h2c to the collector.
+void grpc_export(const State& s, const char* service_method, std::string msg) {
+  grpcx::Target t;
+  t.host = s.grpc_host;
+  t.port = s.grpc_port;
+  t.authority = s.grpc_host + ":" + std::to_string(s.grpc_port);
+  grpcx::unary_call(t, service_method, msg, /*timeout_ms=*/5000);
+  state().delivered.fetch_add(1, std::memory_order_relaxed);
+}
+
 void export_once() {
   State& s = state();
   // ---- spans (chunked: a slow collector must not grow our heap) ----
@@ -203,7 +233,14 @@ void export_once() {
     size_t n = std::min(kMaxSpansPerPost, all.size() - base);
     std::vector<FinishedSpan> batch(all.begin() + static_cast<long>(base),
                                     all.begin() + static_cast<long>(base + n));
-    if (s.use_protobuf) {
+    if (s.transport == Transport::Grpc) {
+      try {
+        grpc_export(s, "/opentelemetry.proto.collector.trace.v1.TraceService/Export",
+                    encode_spans_pb(s, batch));
+      } catch (const std::exception&) { /* collector away; drop batch */ }
+      continue;
+    }
+    if (s.transport == Transport::HttpProtobuf) {
       try {
         post_payload(s.endpoint + "/v1/traces", encode_spans_pb(s, batch),
                      "application/x-protobuf");
@@ -213,8 +250,9 @@ void export_once() {
     jsn::Value spans = jsn::Value::array();
     for (const auto& fs : batch) {
       jsn::Value sp = jsn::Value::object();
-      sp["traceId"] = s.trace_id;
-      sp["spanId"] = rand_hex(8);
+      sp["traceId"] = fs.trace_id;
+      sp["spanId"] = fs.span_id;
+      if (!fs.parent_id.empty()) sp["parentSpanId"] = fs.parent_id;
       sp["name"] = fs.name;
       sp["kind"] = 1;  // SPAN_KIND_INTERNAL
       sp["startTimeUnixNano"] = std::to_string(fs.start_ns);
@@ -237,7 +275,12 @@ void export_once() {
 
   // ---- metrics: the counter registry (monotonic counters + gauges) ----
   auto counters = logx::counters_snapshot();
-  if (!counters.empty() && s.use_protobuf) {
+  if (!counters.empty() && s.transport == Transport::Grpc) {
+    try {
+      grpc_export(s, "/opentelemetry.proto.collector.metrics.v1.MetricsService/Export",
+                  encode_metrics_pb(s, counters, now_unix_ns()));
+    } catch (const std::exception&) { /* collector away */ }
+  } else if (!counters.empty() && s.transport == Transport::HttpProtobuf) {
     try {
       post_payload(s.endpoint + "/v1/metrics",
                    encode_metrics_pb(s, counters, now_unix_ns()),
@@ -293,11 +336,23 @@ void init(const std::string& service_name) {
   while (!s.endpoint.empty() && s.endpoint.back() == '/') s.endpoint.pop_back();
   s.service_name = service_name;
   if (const char* sn = std::getenv("OTEL_SERVICE_NAME"); sn && *sn) s.service_name = sn;
-  s.trace_id = rand_hex(16);
   if (const char* iv = std::getenv("OTEL_METRIC_EXPORT_INTERVAL"); iv && *iv)
     s.interval_ms = std::atoi(iv);
-  if (const char* proto = std::getenv("OTEL_EXPORTER_OTLP_PROTOCOL"); proto && *proto)
-    s.use_protobuf = std::string(proto) == "http/protobuf";
+  // Transport: explicit OTEL_EXPORTER_OTLP_PROTOCOL wins; otherwise a :4317
+  // endpoint gets gRPC (the OTLP default port tonic/the reference target)
+  // and anything else OTLP/HTTP+JSON.
+  auto parsed = http::Url::parse(s.endpoint);
+  if (parsed) {
+    s.grpc_host = parsed->host;
+    s.grpc_port = parsed->port;
+  }
+  const char* proto = std::getenv("OTEL_EXPORTER_OTLP_PROTOCOL");
+  std::string p = proto ? proto : "";
+  if (p == "grpc") s.transport = Transport::Grpc;
+  else if (p == "http/protobuf") s.transport = Transport::HttpProtobuf;
+  else if (p == "http/json") s.transport = Transport::HttpJson;
+  else s.transport = (parsed && parsed->port == 4317) ? Transport::Grpc
+                                                      : Transport::HttpJson;
   s.enabled.store(true);
   s.running.store(true);
   s.exporter = std::thread([&s] {
@@ -328,9 +383,31 @@ void shutdown() {
 
 bool enabled() { return state().enabled.load(std::memory_order_relaxed); }
 
-SpanGuard::SpanGuard(const std::string& name) : name_(name), start_ns_(now_unix_ns()) {}
+SpanContext current_context() {
+  if (!t_ctx.stack.empty()) return {t_ctx.trace_id, t_ctx.stack.back()};
+  if (!t_ctx.installed_parent.empty()) return {t_ctx.trace_id, t_ctx.installed_parent};
+  return {};
+}
+
+SpanGuard::SpanGuard(const std::string& name) : name_(name), start_ns_(now_unix_ns()) {
+  if (!enabled()) return;  // ids stay empty; destructor is a no-op
+  span_id_ = rand_hex(8);
+  if (!t_ctx.stack.empty()) {
+    parent_id_ = t_ctx.stack.back();
+    trace_id_ = t_ctx.trace_id;
+  } else if (!t_ctx.installed_parent.empty()) {
+    parent_id_ = t_ctx.installed_parent;
+    trace_id_ = t_ctx.trace_id;
+  } else {
+    trace_id_ = rand_hex(16);  // root span: fresh trace
+    t_ctx.trace_id = trace_id_;
+  }
+  t_ctx.stack.push_back(span_id_);
+}
 
 SpanGuard::~SpanGuard() {
+  if (span_id_.empty()) return;  // was created before/without otlp::init
+  if (!t_ctx.stack.empty() && t_ctx.stack.back() == span_id_) t_ctx.stack.pop_back();
   if (!enabled()) return;
   State& s = state();
   std::lock_guard<std::mutex> lock(s.mu);
@@ -341,7 +418,22 @@ SpanGuard::~SpanGuard() {
                   s.spans.begin() + static_cast<long>(kMaxBufferedSpans / 10));
     s.dropped_spans.fetch_add(kMaxBufferedSpans / 10, std::memory_order_relaxed);
   }
-  s.spans.push_back({name_, start_ns_, now_unix_ns()});
+  s.spans.push_back({name_, trace_id_, span_id_, parent_id_, start_ns_, now_unix_ns()});
+}
+
+ContextGuard::ContextGuard(const SpanContext& parent) {
+  if (parent.span_id.empty()) return;
+  saved_trace_id_ = t_ctx.trace_id;
+  saved_parent_ = t_ctx.installed_parent;
+  t_ctx.trace_id = parent.trace_id;
+  t_ctx.installed_parent = parent.span_id;
+  installed_ = true;
+}
+
+ContextGuard::~ContextGuard() {
+  if (!installed_) return;
+  t_ctx.trace_id = saved_trace_id_;
+  t_ctx.installed_parent = saved_parent_;
 }
 
 uint64_t delivered_batches() { return state().delivered.load(std::memory_order_relaxed); }
