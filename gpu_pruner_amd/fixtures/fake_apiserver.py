@@ -69,6 +69,12 @@ class FakeApiServer:
         self.throttle_next = 0
         self.retry_after_s = 0
         self.throttled = 0  # how many 429s were actually served
+        # watch support: monotonic resourceVersion + event log
+        self._rv = 1
+        self._events: list[tuple[int, str, str, str, dict]] = []  # (rv, type, kind, ns, obj)
+        self._event_cv = threading.Condition(self._lock)
+        self._closing = False
+        self.watch_requests = 0  # watch connections served (test observability)
 
         fixture = self
 
@@ -117,6 +123,54 @@ class FakeApiServer:
                     return None
                 return kind, ns, name, sub
 
+            def _serve_watch(self, kind, ns, params):
+                """Kubernetes watch: chunked stream of {"type","object"} JSON
+                lines for events after ?resourceVersion, live until
+                timeoutSeconds, closing with a BOOKMARK carrying the latest
+                resourceVersion (allowWatchBookmarks semantics)."""
+                try:
+                    since = int(params.get("resourceVersion", ["0"])[0] or 0)
+                except ValueError:
+                    since = 0
+                timeout_s = float(params.get("timeoutSeconds", ["30"])[0])
+                deadline = time.monotonic() + timeout_s
+                with fixture._lock:
+                    fixture.watch_requests += 1
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Transfer-Encoding", "chunked")
+                self.end_headers()
+
+                def chunk(data: bytes):
+                    self.wfile.write(f"{len(data):x}\r\n".encode() + data + b"\r\n")
+                    self.wfile.flush()
+
+                last = since
+                try:
+                    while not fixture._closing:
+                        with fixture._event_cv:
+                            pending = [e for e in fixture._events
+                                       if e[0] > last and e[2] == kind and e[3] == ns]
+                            if not pending:
+                                remaining = deadline - time.monotonic()
+                                if remaining <= 0:
+                                    break
+                                fixture._event_cv.wait(min(remaining, 0.2))
+                                continue
+                        for rv, etype, _k, _n, obj in pending:
+                            chunk(json.dumps({"type": etype, "object": obj}).encode()
+                                  + b"\n")
+                            last = rv
+                    with fixture._lock:
+                        rv_now = str(fixture._rv)
+                    chunk(json.dumps({"type": "BOOKMARK", "object": {
+                        "kind": kind, "metadata": {"resourceVersion": rv_now}}}).encode()
+                        + b"\n")
+                    self.wfile.write(b"0\r\n\r\n")
+                    self.wfile.flush()
+                except (BrokenPipeError, ConnectionResetError, OSError):
+                    pass  # client went away mid-stream
+
             def do_GET(self):
                 with fixture._lock:
                     fixture.requests.append(("GET", self.path))
@@ -145,10 +199,17 @@ class FakeApiServer:
                     return self._send(404, {"kind": "Status", "code": 404})
                 kind, ns, name, _sub = r
                 if name is None:
+                    import urllib.parse as _up
+                    params = _up.parse_qs(query)
+                    if params.get("watch", ["false"])[0] == "true":
+                        return self._serve_watch(kind, ns, params)
                     with fixture._lock:
                         items = [copy.deepcopy(o) for (k, n, _), o in fixture.objects.items()
                                  if k == kind and n == ns]
-                    return self._send(200, {"kind": kind + "List", "items": items})
+                        rv = str(fixture._rv)
+                    return self._send(200, {"kind": kind + "List",
+                                            "metadata": {"resourceVersion": rv},
+                                            "items": items})
                 with fixture._lock:
                     obj = fixture.objects.get((kind, ns, name))
                     if obj is None:
@@ -182,6 +243,7 @@ class FakeApiServer:
                             return self._send(400, {"kind": "Status", "code": 400,
                                                     "message": "scale patch needs spec.replicas"})
                         obj.setdefault("spec", {})["replicas"] = replicas
+                        fixture._record_locked("MODIFIED", kind, obj)
                         scale = {
                             "kind": "Scale", "apiVersion": "autoscaling/v1",
                             "metadata": {"name": name, "namespace": ns},
@@ -189,6 +251,7 @@ class FakeApiServer:
                         }
                         return self._send(200, scale)
                     _merge_patch(obj, patch)
+                    fixture._record_locked("MODIFIED", kind, obj)
                     return self._send(200, copy.deepcopy(obj))
 
             def do_POST(self):
@@ -213,6 +276,7 @@ class FakeApiServer:
                 name = obj.get("metadata", {}).get("name", "")
                 with fixture._lock:
                     fixture.objects[(kind, ns, name)] = obj
+                    fixture._record_locked("ADDED", kind, obj)
                 return self._send(201, obj)
 
         self._server = ThreadingHTTPServer((host, port), Handler)
@@ -236,6 +300,9 @@ class FakeApiServer:
         return self
 
     def stop(self):
+        with self._event_cv:
+            self._closing = True
+            self._event_cv.notify_all()
         self._server.shutdown()
         self._server.server_close()
 
@@ -252,10 +319,27 @@ class FakeApiServer:
         self.stop()
 
     # -- object builders -----------------------------------------------------
+    def _record_locked(self, etype: str, kind: str, obj: dict):
+        """Bump the global resourceVersion, stamp the object, log the event.
+        Caller holds self._lock."""
+        self._rv += 1
+        obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
+        self._events.append((self._rv, etype, kind,
+                             obj["metadata"].get("namespace", ""), copy.deepcopy(obj)))
+        self._event_cv.notify_all()
+
     def put(self, kind: str, obj: dict):
         meta = obj["metadata"]
         with self._lock:
             self.objects[(kind, meta["namespace"], meta["name"])] = obj
+            self._record_locked("ADDED", kind, obj)
+        return obj
+
+    def delete_object(self, kind: str, ns: str, name: str):
+        with self._lock:
+            obj = self.objects.pop((kind, ns, name), None)
+            if obj is not None:
+                self._record_locked("DELETED", kind, obj)
         return obj
 
     def get(self, kind: str, ns: str, name: str) -> dict | None:

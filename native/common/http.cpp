@@ -339,11 +339,103 @@ private:
     }
   }
 
+public:
+  // Abort a blocked read from another thread (watch-stream shutdown).
+  void shutdown_socket() {
+    broken_ = true;
+    if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
+  }
+
   ClientOptions opts_;
   int fd_ = -1;
   SSL* ssl_ = nullptr;
   bool broken_ = false;
 };
+
+namespace {
+
+// Serialize a request against `base` with `default_headers` (shared by the
+// buffered and streaming paths).
+std::string serialize_request(
+    const Url& base, const std::vector<std::pair<std::string, std::string>>& default_headers,
+    const Request& req) {
+  std::string host_hdr =
+      base.scheme == "unix" ? "localhost" : base.host + ":" + std::to_string(base.port);
+  std::string out;
+  out.reserve(512 + req.body.size());
+  out += req.method + " " + (req.path.empty() ? "/" : req.path) + " HTTP/1.1\r\n";
+  out += "Host: " + host_hdr + "\r\n";
+  bool have_accept = false, have_ct = false;
+  auto append_hdr = [&](const std::string& k, const std::string& v) {
+    std::string lk = strutil::lower(k);
+    if (lk == "accept") have_accept = true;
+    if (lk == "content-type") have_ct = true;
+    out += k + ": " + v + "\r\n";
+  };
+  for (const auto& [k, v] : default_headers) {
+    bool overridden = false;
+    for (const auto& [rk, rv] : req.headers)
+      if (strutil::lower(rk) == strutil::lower(k)) overridden = true;
+    if (!overridden) append_hdr(k, v);
+  }
+  for (const auto& [k, v] : req.headers) append_hdr(k, v);
+  if (!have_accept) out += "Accept: application/json\r\n";
+  if (!req.body.empty() && !have_ct) out += "Content-Type: application/json\r\n";
+  if (!req.body.empty() || req.method == "POST" || req.method == "PUT" || req.method == "PATCH")
+    out += "Content-Length: " + std::to_string(req.body.size()) + "\r\n";
+  out += "Connection: keep-alive\r\n\r\n";
+  out += req.body;
+  return out;
+}
+
+// Read status line + headers (skipping interim 1xx). Leaves any body bytes
+// already received in *leftover.
+void read_response_head(Connection& c, Response* resp, std::string* leftover) {
+  std::string buf;
+  std::string rest;
+  while (true) {
+    char tmp[8192];
+    while (buf.find("\r\n\r\n") == std::string::npos) {
+      size_t r = c.read_some(tmp, sizeof tmp);
+      if (r == 0) {
+        c.mark_broken();
+        throw Error("connection closed before response headers");
+      }
+      buf.append(tmp, r);
+      if (buf.size() > (1u << 20)) {
+        c.mark_broken();
+        throw Error("response headers too large");
+      }
+    }
+    *resp = Response{};
+    size_t hdr_end = buf.find("\r\n\r\n");
+    std::string head = buf.substr(0, hdr_end);
+    rest = buf.substr(hdr_end + 4);
+    auto lines = strutil::split(head, '\n');
+    if (lines.empty()) throw Error("malformed response");
+    {
+      std::string status_line = strutil::trim(lines[0]);
+      size_t sp1 = status_line.find(' ');
+      if (sp1 == std::string::npos) throw Error("malformed status line: " + status_line);
+      resp->status = std::atoi(status_line.c_str() + sp1 + 1);
+    }
+    for (size_t i = 1; i < lines.size(); i++) {
+      std::string line = strutil::trim(lines[i]);
+      size_t colon = line.find(':');
+      if (colon == std::string::npos) continue;
+      resp->headers[strutil::lower(line.substr(0, colon))] =
+          strutil::trim(line.substr(colon + 1));
+    }
+    if (resp->status >= 100 && resp->status < 200) {
+      buf = rest;
+      continue;
+    }
+    break;
+  }
+  *leftover = std::move(rest);
+}
+
+}  // namespace
 
 // ------------------------------ Client ---------------------------------------
 
@@ -447,82 +539,12 @@ Response Client::request(const Request& req) {
 }
 
 Response Client::do_request_on(Connection& c, const Request& req) {
-  // (interim 1xx responses are skipped below without re-sending the request)
-  std::string host_hdr =
-      base_.scheme == "unix" ? "localhost" : base_.host + ":" + std::to_string(base_.port);
-  std::string out;
-  out.reserve(512 + req.body.size());
-  out += req.method + " " + (req.path.empty() ? "/" : req.path) + " HTTP/1.1\r\n";
-  out += "Host: " + host_hdr + "\r\n";
-  bool have_accept = false, have_ct = false;
-  auto append_hdr = [&](const std::string& k, const std::string& v) {
-    std::string lk = strutil::lower(k);
-    if (lk == "accept") have_accept = true;
-    if (lk == "content-type") have_ct = true;
-    out += k + ": " + v + "\r\n";
-  };
-  for (const auto& [k, v] : default_headers_) {
-    bool overridden = false;
-    for (const auto& [rk, rv] : req.headers)
-      if (strutil::lower(rk) == strutil::lower(k)) overridden = true;
-    if (!overridden) append_hdr(k, v);
-  }
-  for (const auto& [k, v] : req.headers) append_hdr(k, v);
-  if (!have_accept) out += "Accept: application/json\r\n";
-  if (!req.body.empty() && !have_ct) out += "Content-Type: application/json\r\n";
-  if (!req.body.empty() || req.method == "POST" || req.method == "PUT" || req.method == "PATCH")
-    out += "Content-Length: " + std::to_string(req.body.size()) + "\r\n";
-  out += "Connection: keep-alive\r\n\r\n";
-  out += req.body;
-
+  std::string out = serialize_request(base_, default_headers_, req);
   c.write_all(out.data(), out.size());
 
-  // ---- read status line + headers (skipping any interim 1xx) ----
-  std::string buf;
   Response resp;
   std::string rest;
-  while (true) {
-    char tmp[8192];
-    while (buf.find("\r\n\r\n") == std::string::npos) {
-      size_t r = c.read_some(tmp, sizeof tmp);
-      if (r == 0) {
-        c.mark_broken();
-        throw Error("connection closed before response headers");
-      }
-      buf.append(tmp, r);
-      if (buf.size() > (1u << 20)) {
-        c.mark_broken();
-        throw Error("response headers too large");
-      }
-    }
-    resp = Response{};
-    size_t hdr_end = buf.find("\r\n\r\n");
-    std::string head = buf.substr(0, hdr_end);
-    rest = buf.substr(hdr_end + 4);
-    auto lines = strutil::split(head, '\n');
-    if (lines.empty()) throw Error("malformed response");
-    {
-      std::string status_line = strutil::trim(lines[0]);
-      // HTTP/1.1 200 OK
-      size_t sp1 = status_line.find(' ');
-      if (sp1 == std::string::npos) throw Error("malformed status line: " + status_line);
-      resp.status = std::atoi(status_line.c_str() + sp1 + 1);
-    }
-    for (size_t i = 1; i < lines.size(); i++) {
-      std::string line = strutil::trim(lines[i]);
-      size_t colon = line.find(':');
-      if (colon == std::string::npos) continue;
-      resp.headers[strutil::lower(line.substr(0, colon))] =
-          strutil::trim(line.substr(colon + 1));
-    }
-    if (resp.status >= 100 && resp.status < 200) {
-      // interim response (100-continue etc.): it has no body; keep reading
-      // for the real one without re-sending anything
-      buf = rest;
-      continue;
-    }
-    break;
-  }
+  read_response_head(c, &resp, &rest);
 
   bool keep_alive = true;
   {
@@ -626,6 +648,112 @@ Response Client::patch(const std::string& path, const std::string& body,
   r.headers = headers;
   r.headers.emplace_back("Content-Type", content_type);
   return request(r);
+}
+
+// ---------------------------- BodyStream -------------------------------------
+
+BodyStream::BodyStream(std::unique_ptr<Connection> conn, int status,
+                       std::map<std::string, std::string> headers, std::string initial,
+                       bool chunked)
+    : conn_(std::move(conn)), status_(status), headers_(std::move(headers)),
+      raw_(std::move(initial)), chunked_(chunked) {}
+
+BodyStream::~BodyStream() = default;
+
+void BodyStream::shutdown() {
+  if (conn_) conn_->shutdown_socket();
+}
+
+bool BodyStream::fill() {
+  // move decoded bytes from raw_ (and the socket) into buf_
+  while (true) {
+    if (!chunked_) {
+      if (!raw_.empty()) {
+        buf_ += raw_;
+        raw_.clear();
+        return true;
+      }
+    } else {
+      // decode whatever complete chunk data we have
+      bool progressed = false;
+      while (true) {
+        if (chunk_remaining_ > 0) {
+          size_t take = std::min(chunk_remaining_, raw_.size());
+          if (take == 0) break;
+          buf_.append(raw_, 0, take);
+          raw_.erase(0, take);
+          chunk_remaining_ -= take;
+          progressed = true;
+          if (chunk_remaining_ == 0) {
+            // consume the trailing CRLF when it arrives
+            if (raw_.size() >= 2) raw_.erase(0, 2);
+            else chunk_remaining_ = 0;  // CRLF split across reads: handled below
+          }
+          continue;
+        }
+        // skip a stray CRLF left from a chunk boundary
+        while (!raw_.empty() && (raw_[0] == '\r' || raw_[0] == '\n')) raw_.erase(0, 1);
+        size_t crlf = raw_.find("\r\n");
+        if (crlf == std::string::npos) break;  // need more bytes for the size line
+        size_t len = std::strtoul(raw_.substr(0, crlf).c_str(), nullptr, 16);
+        raw_.erase(0, crlf + 2);
+        if (len == 0) {  // final chunk
+          eof_ = true;
+          return progressed;
+        }
+        chunk_remaining_ = len;
+      }
+      if (progressed) return true;
+    }
+    if (eof_) return false;
+    char tmp[16384];
+    size_t r;
+    try {
+      r = conn_->read_some(tmp, sizeof tmp);
+    } catch (const Error&) {
+      eof_ = true;
+      throw;
+    }
+    if (r == 0) {
+      eof_ = true;
+      return !raw_.empty() || !buf_.empty();
+    }
+    raw_.append(tmp, r);
+  }
+}
+
+bool BodyStream::read_line(std::string* line) {
+  while (true) {
+    size_t nl = buf_.find('\n');
+    if (nl != std::string::npos) {
+      *line = buf_.substr(0, nl);
+      if (!line->empty() && line->back() == '\r') line->pop_back();
+      buf_.erase(0, nl + 1);
+      return true;
+    }
+    if (eof_) {
+      if (buf_.empty()) return false;
+      *line = std::move(buf_);  // unterminated trailing line
+      buf_.clear();
+      return true;
+    }
+    if (!fill() && eof_ && buf_.empty()) return false;
+  }
+}
+
+std::unique_ptr<BodyStream> Client::open_stream(const Request& req) {
+  auto conn = std::make_unique<Connection>(base_, opts_, static_cast<SSL_CTX*>(ssl_ctx_));
+  std::string out = serialize_request(base_, default_headers_, req);
+  conn->write_all(out.data(), out.size());
+  Response head;
+  std::string leftover;
+  read_response_head(*conn, &head, &leftover);
+  bool chunked = false;
+  if (auto it = head.headers.find("transfer-encoding"); it != head.headers.end())
+    chunked = strutil::lower(it->second).find("chunked") != std::string::npos;
+  return std::make_unique<BodyStream>(std::move(conn), head.status,
+                                      std::move(head.headers), std::move(leftover),
+                                      chunked);
 }
 
 Response fetch(const std::string& url, const Request& req, const ClientOptions& opts) {

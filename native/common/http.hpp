@@ -66,6 +66,41 @@ struct ClientOptions {
 
 class Connection;  // opaque: one TCP/TLS (or unix) stream
 
+// Incrementally-read response body (Kubernetes watch streams: chunked JSON
+// events that arrive over minutes). Owns its connection; the connection is
+// NOT returned to the pool (watch streams are not reusable).
+class BodyStream {
+public:
+  BodyStream(std::unique_ptr<Connection> conn, int status,
+             std::map<std::string, std::string> headers, std::string initial,
+             bool chunked);
+  ~BodyStream();
+
+  int status() const { return status_; }
+  const std::map<std::string, std::string>& headers() const { return headers_; }
+
+  // Next newline-terminated line of the decoded body (without the '\n').
+  // Returns false on orderly end of stream. Throws Error on transport
+  // errors/timeouts (io_timeout_ms of the owning client's options).
+  bool read_line(std::string* line);
+
+  // Abort from another thread: shuts the socket down so a blocked
+  // read_line returns/throws promptly.
+  void shutdown();
+
+private:
+  bool fill();  // read more decoded bytes into buf_; false on EOF
+
+  std::unique_ptr<Connection> conn_;
+  int status_;
+  std::map<std::string, std::string> headers_;
+  std::string raw_;      // undecoded (possibly chunked) bytes
+  std::string buf_;      // decoded body bytes not yet consumed
+  bool chunked_;
+  bool eof_ = false;
+  size_t chunk_remaining_ = 0;  // bytes left in the current chunk's data
+};
+
 // Thread-safe HTTP client for one origin (scheme+host+port). Connections are
 // pooled and reused across requests; a request that finds the pooled
 // connection stale (server closed keep-alive) is retried once on a fresh one.
@@ -79,6 +114,10 @@ public:
 
   // `path` overrides base.path; headers are appended to defaults.
   Response request(const Request& req);
+
+  // Open a streaming request (Kubernetes watch): returns after status +
+  // headers arrive; body bytes are pulled incrementally via BodyStream.
+  std::unique_ptr<BodyStream> open_stream(const Request& req);
 
   // convenience
   Response get(const std::string& path,

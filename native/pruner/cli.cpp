@@ -44,7 +44,8 @@ OPTIONS:
                                   failures [default: 5]
       --metrics-port <PORT>       serve daemon self-metrics + /healthz on
                                   this port (0 = disabled) [default: 0]
-      --eval-strategy <S>         candidate fetch strategy: get (per-object
+      --eval-strategy <S>         candidate fetch strategy: watch (persistent
+                                  informers, delta traffic), get (per-object
                                   GETs, reference-equivalent) | list
                                   (namespace LISTs) | auto [default: auto]
   -h, --help                      print this help
@@ -180,7 +181,8 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
         if (m == "get") c.eval_strategy = EvalStrategy::PerPodGet;
         else if (m == "list") c.eval_strategy = EvalStrategy::NamespaceList;
         else if (m == "auto") c.eval_strategy = EvalStrategy::Auto;
-        else return fail("invalid --eval-strategy (expected get|list|auto): " + val);
+        else if (m == "watch") c.eval_strategy = EvalStrategy::Watch;
+        else return fail("invalid --eval-strategy (expected get|list|auto|watch): " + val);
       }
     } catch (const std::exception&) {
       return fail("invalid value for --" + name + ": " + val);

@@ -11,6 +11,7 @@
 #include <vector>
 
 #include "../common/http_server.hpp"
+#include "informer.hpp"
 #include "../common/log.hpp"
 #include "../common/queue.hpp"
 #include "engine.hpp"
@@ -172,6 +173,7 @@ int run_daemon(const Config& cfg) {
   queue.close();  // producer done: consumers drain and exit
   for (auto& c : consumers) c.join();
   if (metrics_server) metrics_server->stop();
+  InformerRegistry::global().stop_all();  // close watch streams (if any)
   return exit_code.load();
 }
 

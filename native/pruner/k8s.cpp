@@ -359,6 +359,15 @@ jsn::Value KubeClient::patch_scale(Kind kind, const std::string& ns, const std::
   return merge_patch(object_path(kind, ns, name) + "/scale", patch);
 }
 
+std::unique_ptr<http::BodyStream> KubeClient::open_stream(const std::string& path) {
+  http::Request r;
+  r.method = "GET";
+  r.path = path;
+  std::string token = bearer();
+  if (!token.empty()) r.headers.emplace_back("Authorization", "Bearer " + token);
+  return http_->open_stream(r);
+}
+
 jsn::Value KubeClient::create(const std::string& collection, const jsn::Value& obj) {
   http::Request r;
   r.method = "POST";

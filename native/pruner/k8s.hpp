@@ -90,6 +90,9 @@ public:
 
   jsn::Value create(const std::string& collection_path, const jsn::Value& obj);
 
+  // Streaming GET (Kubernetes watch): returns once status+headers arrive.
+  std::unique_ptr<http::BodyStream> open_stream(const std::string& path);
+
   const KubeConfig& config() const { return cfg_; }
 
 private:

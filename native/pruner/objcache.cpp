@@ -22,6 +22,29 @@ struct ListJob {
 void ObjectCache::prefetch(const std::map<std::string, int>& ns_counts, int concurrency) {
   if (strategy_ == EvalStrategy::PerPodGet) return;
 
+  if (strategy_ == EvalStrategy::Watch) {
+    // Persistent informers: created on the first tick that sees a namespace
+    // (each LISTs once in its own thread), free from the second tick on.
+    auto& reg = InformerRegistry::global();
+    for (const auto& [ns, count] : ns_counts) {
+      (void)count;
+      informers_[{ns, "Pod"}] = &reg.get_or_create(
+          kube_.config(), ns, "Pod",
+          "/api/v1/namespaces/" + strutil::url_encode(ns) + "/pods");
+      for (Kind k : {Kind::ReplicaSet, Kind::Deployment, Kind::StatefulSet, Kind::Notebook,
+                     Kind::InferenceService})
+        informers_[{ns, kind_name(k)}] =
+            &reg.get_or_create(kube_.config(), ns, kind_name(k), collection_path(k, ns));
+    }
+    // wait for initial syncs (in parallel across informer threads); a
+    // not-yet-synced informer falls back to GETs this tick
+    for (auto& [key, inf] : informers_)
+      if (!inf->wait_synced(10000))
+        LOGW(TARGET, "informer for " + key.first + "/" + key.second +
+                         " not synced; falling back to GETs this tick");
+    return;
+  }
+
   std::vector<std::string> namespaces;
   for (const auto& [ns, count] : ns_counts) {
     if (strategy_ == EvalStrategy::NamespaceList || count >= auto_threshold_)
@@ -66,6 +89,9 @@ void ObjectCache::prefetch(const std::map<std::string, int>& ns_counts, int conc
 
 std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
                                                const std::string& name) {
+  if (auto iit = informers_.find({ns, "Pod"}); iit != informers_.end()) {
+    if (iit->second->synced()) return iit->second->get(name);
+  }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     if (auto kit = nit->second.by_kind.find("Pod"); kit != nit->second.by_kind.end()) {
       auto oit = kit->second.find(name);
@@ -78,6 +104,9 @@ std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
 
 std::optional<jsn::Value> ObjectCache::get_object(Kind kind, const std::string& ns,
                                                   const std::string& name) {
+  if (auto iit = informers_.find({ns, kind_name(kind)}); iit != informers_.end()) {
+    if (iit->second->synced()) return iit->second->get(name);
+  }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     auto kit = nit->second.by_kind.find(kind_name(kind));
     if (kit != nit->second.by_kind.end()) {

@@ -8,10 +8,14 @@
 // decision semantics (a pod absent from a fresh LIST is exactly as gone as
 // a 404), two orders of magnitude fewer round-trips at 1000-pod scale.
 //
-//   get  — always per-object GETs (reference-equivalent behavior)
-//   list — always prefetch collections for candidate namespaces
-//   auto — LIST namespaces with >= threshold candidates, GETs elsewhere
-//          (big LISTs are not free on huge namespaces with few candidates)
+//   get   — always per-object GETs (reference-equivalent behavior)
+//   list  — always prefetch collections for candidate namespaces
+//   auto  — LIST namespaces with >= threshold candidates, GETs elsewhere
+//           (big LISTs are not free on huge namespaces with few candidates)
+//   watch — read through persistent informers (informer.hpp): one LIST when
+//           a collection is first seen, then watch-event deltas keep the
+//           store current across ticks — steady-state per-tick apiserver
+//           traffic is O(changes), not O(objects) (daemon mode)
 #pragma once
 
 #include <map>
@@ -19,12 +23,13 @@
 #include <set>
 #include <string>
 
+#include "informer.hpp"
 #include "k8s.hpp"
 #include "resources.hpp"
 
 namespace pruner {
 
-enum class EvalStrategy { PerPodGet, NamespaceList, Auto };
+enum class EvalStrategy { PerPodGet, NamespaceList, Auto, Watch };
 
 class ObjectCache {
 public:
@@ -55,6 +60,8 @@ private:
   EvalStrategy strategy_;
   int auto_threshold_;
   std::map<std::string, NsCache> cache_;  // namespace → cache
+  // watch strategy: per-(ns, kind) informer handles for this tick's lookups
+  std::map<std::pair<std::string, std::string>, Informer*> informers_;
   size_t lists_issued_ = 0;
 };
 

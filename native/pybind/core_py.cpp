@@ -21,6 +21,7 @@
 #include "../pruner/resources.hpp"
 #include "../common/http.hpp"
 #include "../common/miniyaml.hpp"
+#include "../pruner/informer.hpp"
 #include "../pruner/otlp.hpp"
 #include "../pruner/synthbench.hpp"
 
@@ -66,9 +67,10 @@ Config config_from_json(const std::string& json_cfg) {
     c.run_mode = RunMode::ScaleDown;
   c.enabled_resources = v.get("enabled_resources").as_string_or("drsin");
   std::string strat = v.get("eval_strategy").as_string_or("auto");
-  c.eval_strategy = strat == "get"    ? EvalStrategy::PerPodGet
-                    : strat == "list" ? EvalStrategy::NamespaceList
-                                      : EvalStrategy::Auto;
+  c.eval_strategy = strat == "get"     ? EvalStrategy::PerPodGet
+                    : strat == "list"  ? EvalStrategy::NamespaceList
+                    : strat == "watch" ? EvalStrategy::Watch
+                                       : EvalStrategy::Auto;
   if (v.get("prometheus_url").is_string())
     c.prometheus_url = v.get("prometheus_url").as_string();
   return c;
@@ -257,6 +259,13 @@ PYBIND11_MODULE(_pruner_core, m) {
         "scale-down mode) inline actuation of every selected root");
 
   m.def("get_prometheus_token", [] { return get_prometheus_token(); });
+
+  m.def("informers_reset",
+        [] {
+          py::gil_scoped_release nogil;
+          InformerRegistry::global().stop_all();
+        },
+        "Stop and drop all persistent watch informers (test isolation)");
 
   m.def("resolve_kube_config", [] {
     KubeConfig cfg = KubeConfig::resolve();
