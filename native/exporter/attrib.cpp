@@ -201,19 +201,49 @@ namespace {
 
 // AMD device-plugin device IDs vary by plugin version/config; match a
 // reported id against every stable identity the sampler knows for a device.
+// Known forms (ROCm/k8s-device-plugin keys its devices by the PCI address
+// from /sys/module/amdgpu/drivers/pci:amdgpu; other stacks report KFD
+// gpu_id, the 64-bit unique id, DRM node names, or a bare index):
+//   "0000:23:00.0" / "0000:23:00"  — PCI BDF, with/without function
+//   "23:00.0"                      — BDF without the PCI domain
+//   "56525"                        — KFD topology gpu_id
+//   "32da0b77724e0fe2" / "0x32da…" — unique id (zero-padding optional)
+//   "card2" / "renderD152" / "/dev/dri/renderD152"
+//   "0"                            — device index
 bool device_id_matches(const std::string& raw_id, const DeviceSample& d) {
   std::string id = strutil::lower(strutil::trim(raw_id));
   if (id.empty()) return false;
   if (id == std::to_string(d.kfd_gpu_id)) return true;
   std::string uid = strutil::lower(d.unique_id);
-  if (!uid.empty() && (id == uid || id == "0x" + uid)) return true;
+  if (!uid.empty()) {
+    if (id == uid || id == "0x" + uid) return true;
+    // unpadded hex: compare with leading zeros stripped from both sides
+    auto strip = [](std::string s) {
+      if (s.rfind("0x", 0) == 0) s.erase(0, 2);
+      size_t nz = s.find_first_not_of('0');
+      return nz == std::string::npos ? std::string("0") : s.substr(nz);
+    };
+    if (strip(id) == strip(uid) && id.find_first_not_of("0123456789abcdefx") ==
+                                       std::string::npos)
+      return true;
+  }
   if (id == "renderd" + std::to_string(d.drm_render_minor)) return true;
   if (id == "/dev/dri/renderd" + std::to_string(d.drm_render_minor)) return true;
   // card index convention: render minor 128+N ↔ cardN
   if (d.drm_render_minor >= 128 &&
       id == "card" + std::to_string(d.drm_render_minor - 128))
     return true;
-  if (id == strutil::lower(d.pci_bdf)) return true;
+  std::string bdf = strutil::lower(d.pci_bdf);
+  if (!bdf.empty()) {
+    if (id == bdf) return true;
+    // without the function suffix ("0000:23:00")
+    if (size_t dot = bdf.rfind('.'); dot != std::string::npos && id == bdf.substr(0, dot))
+      return true;
+    // without the PCI domain ("23:00.0")
+    if (size_t colon = bdf.find(':'); colon != std::string::npos &&
+                                      id == bdf.substr(colon + 1))
+      return true;
+  }
   if (id == std::to_string(d.index)) return true;
   return false;
 }

@@ -143,3 +143,80 @@ def test_resolve_full_kfd_fallback_without_socket(gpumon, tmp_path, monkeypatch)
     out = a.resolve_full(json.dumps(SAMPLES))
     assert out[0]["pod"] == "kfd-pod"
     assert 1 not in out
+
+
+# ---- AMD device-plugin ID-form fixtures ---------------------
+#
+# ROCm/k8s-device-plugin keys its pluginapi.Device IDs by the PCI address it
+# discovers under /sys/module/amdgpu/drivers/pci:amdgpu; other stacks (GPU
+# operator builds, forks) have shipped KFD gpu_ids, 64-bit unique ids, DRM
+# node names, or bare indices. Each known form must attribute the same GPU.
+
+DEVICE0 = {"index": 0, "kfd_gpu_id": 56525, "unique_id": "0012345678abcdef",
+           "drm_render_minor": 152, "pci_bdf": "0000:23:00.0"}
+
+
+@pytest.mark.parametrize("device_id", [
+    "0000:23:00.0",          # PCI BDF (ROCm k8s-device-plugin's native form)
+    "0000:23:00",            # BDF without function
+    "23:00.0",               # BDF without PCI domain
+    "0000:23:00.0".upper(),  # case-insensitive
+    "56525",                 # KFD topology gpu_id
+    "0012345678abcdef",      # unique id, padded
+    "0x0012345678abcdef",    # unique id, 0x-prefixed
+    "0x12345678abcdef",      # unique id, unpadded
+    "12345678abcdef",        # unique id, unpadded, no prefix
+    "renderD152",            # DRM render node name
+    "/dev/dri/renderD152",   # DRM render node path
+    "card24",                # card index (renderD 128+N ↔ cardN)
+    "0",                     # device index
+])
+def test_device_id_forms_attribute(gpumon, tmp_path, monkeypatch, device_id):
+    from gpu_pruner_amd.fixtures.fake_podresources import FakePodResources
+
+    entries = [{
+        "pod": "train-0", "namespace": "ml",
+        "containers": [{"name": "worker",
+                        "devices": [{"resource_name": "amd.com/gpu",
+                                     "device_ids": [device_id]}]}],
+    }]
+    sock = str(tmp_path / "kubelet.sock")
+    monkeypatch.setenv("GPU_EXPORTER_PODRESOURCES_SOCKET", sock)
+    monkeypatch.setenv("GPU_EXPORTER_SYSFS_ROOT", str(tmp_path))
+    monkeypatch.setenv("GPU_EXPORTER_PROCFS_ROOT", str(tmp_path))
+    monkeypatch.delenv("GPU_EXPORTER_POD_MAP_FILE", raising=False)
+    monkeypatch.delenv("GPU_PRUNER_K8S_URL", raising=False)
+    monkeypatch.delenv("KUBERNETES_SERVICE_HOST", raising=False)
+    with FakePodResources(sock, entries):
+        a = gpumon.Attributor()
+        out = a.resolve_full(json.dumps([DEVICE0]))
+    assert out.get(0) == {"pod": "train-0", "namespace": "ml",
+                          "container": "worker"}, device_id
+
+
+@pytest.mark.parametrize("bogus_id", [
+    "0000:99:00.0",     # different BDF
+    "99999",            # unknown gpu_id
+    "renderD200",       # other render node
+    "gpu-not-a-form",   # garbage
+])
+def test_non_matching_ids_do_not_attribute(gpumon, tmp_path, monkeypatch, bogus_id):
+    from gpu_pruner_amd.fixtures.fake_podresources import FakePodResources
+
+    entries = [{
+        "pod": "train-0", "namespace": "ml",
+        "containers": [{"name": "worker",
+                        "devices": [{"resource_name": "amd.com/gpu",
+                                     "device_ids": [bogus_id]}]}],
+    }]
+    sock = str(tmp_path / "kubelet.sock")
+    monkeypatch.setenv("GPU_EXPORTER_PODRESOURCES_SOCKET", sock)
+    monkeypatch.setenv("GPU_EXPORTER_SYSFS_ROOT", str(tmp_path))
+    monkeypatch.setenv("GPU_EXPORTER_PROCFS_ROOT", str(tmp_path))
+    monkeypatch.delenv("GPU_EXPORTER_POD_MAP_FILE", raising=False)
+    monkeypatch.delenv("GPU_PRUNER_K8S_URL", raising=False)
+    monkeypatch.delenv("KUBERNETES_SERVICE_HOST", raising=False)
+    with FakePodResources(sock, entries):
+        a = gpumon.Attributor()
+        out = a.resolve_full(json.dumps([DEVICE0]))
+    assert 0 not in out, (bogus_id, out)
