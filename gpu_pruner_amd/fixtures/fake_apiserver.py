@@ -95,6 +95,18 @@ class FakeApiServer:
                 self.end_headers()
                 self.wfile.write(body)
 
+            def _status(self, code, reason, message="", details=None):
+                """Full metav1.Status shape, as the real apiserver returns it
+                (API conventions: status=Failure, reason, code, details)."""
+                body = {"kind": "Status", "apiVersion": "v1",
+                        "metadata": {}, "status": "Failure",
+                        "reason": reason, "code": code}
+                if message:
+                    body["message"] = message
+                if details:
+                    body["details"] = details
+                return body
+
             def _maybe_throttle(self) -> bool:
                 with fixture._lock:
                     if fixture.throttle_next <= 0:
@@ -102,8 +114,8 @@ class FakeApiServer:
                     fixture.throttle_next -= 1
                     fixture.throttled += 1
                     ra = fixture.retry_after_s
-                self._send(429, {"kind": "Status", "code": 429,
-                                 "reason": "TooManyRequests"},
+                self._send(429, self._status(429, "TooManyRequests",
+                                             "Too many requests, please try again later."),
                            headers={"Retry-After": str(ra)})
                 return True
 
@@ -179,7 +191,7 @@ class FakeApiServer:
                 if self._maybe_throttle():
                     return
                 if not self._auth_ok():
-                    return self._send(401, {"kind": "Status", "code": 401})
+                    return self._send(401, self._status(401, "Unauthorized", "Unauthorized"))
                 # cluster-scope pod list (used by the exporter's attribution
                 # cache), with optional spec.nodeName fieldSelector
                 path_only, _, query = self.path.partition("?")
@@ -196,7 +208,8 @@ class FakeApiServer:
                     return self._send(200, {"kind": "PodList", "items": items})
                 r = self._route()
                 if r is None:
-                    return self._send(404, {"kind": "Status", "code": 404})
+                    return self._send(404, self._status(404, "NotFound",
+                                                        "the server could not find the requested resource"))
                 kind, ns, name, _sub = r
                 if name is None:
                     import urllib.parse as _up
@@ -213,8 +226,10 @@ class FakeApiServer:
                 with fixture._lock:
                     obj = fixture.objects.get((kind, ns, name))
                     if obj is None:
-                        return self._send(404, {"kind": "Status", "code": 404,
-                                                "message": f"{kind} {ns}/{name} not found"})
+                        return self._send(404, self._status(
+                            404, "NotFound",
+                            f'{kind.lower()}s "{name}" not found',
+                            details={"name": name, "kind": kind.lower() + "s"}))
                     return self._send(200, copy.deepcopy(obj))
 
             def do_PATCH(self):
@@ -225,29 +240,39 @@ class FakeApiServer:
                 if self._maybe_throttle():
                     return
                 if not self._auth_ok():
-                    return self._send(401, {"kind": "Status", "code": 401})
+                    return self._send(401, self._status(401, "Unauthorized", "Unauthorized"))
                 length = int(self.headers.get("Content-Length", "0"))
                 patch = json.loads(self.rfile.read(length) or b"{}")
                 r = self._route()
                 if r is None or r[2] is None:
-                    return self._send(404, {"kind": "Status", "code": 404})
+                    return self._send(404, self._status(404, "NotFound",
+                                                        "the server could not find the requested resource"))
                 kind, ns, name, sub = r
                 with fixture._lock:
                     obj = fixture.objects.get((kind, ns, name))
                     if obj is None:
-                        return self._send(404, {"kind": "Status", "code": 404})
+                        return self._send(404, self._status(
+                            404, "NotFound", f'{kind.lower()}s "{name}" not found',
+                            details={"name": name, "kind": kind.lower() + "s"}))
                     if sub == "scale":
-                        # /scale only understands spec.replicas
+                        # /scale understands only spec.replicas; a merge patch
+                        # without it is a no-op returning the current Scale
+                        # (real-apiserver semantics)
                         replicas = (patch.get("spec") or {}).get("replicas")
-                        if replicas is None:
-                            return self._send(400, {"kind": "Status", "code": 400,
-                                                    "message": "scale patch needs spec.replicas"})
-                        obj.setdefault("spec", {})["replicas"] = replicas
-                        fixture._record_locked("MODIFIED", kind, obj)
+                        if replicas is not None:
+                            obj.setdefault("spec", {})["replicas"] = replicas
+                            fixture._record_locked("MODIFIED", kind, obj)
+                        current = obj.get("spec", {}).get("replicas", 0)
                         scale = {
                             "kind": "Scale", "apiVersion": "autoscaling/v1",
-                            "metadata": {"name": name, "namespace": ns},
-                            "spec": {"replicas": replicas},
+                            "metadata": {
+                                "name": name, "namespace": ns,
+                                "uid": obj.get("metadata", {}).get("uid", ""),
+                                "resourceVersion": obj.get("metadata", {}).get(
+                                    "resourceVersion", "1"),
+                            },
+                            "spec": {"replicas": current},
+                            "status": {"replicas": current},
                         }
                         return self._send(200, scale)
                     _merge_patch(obj, patch)
@@ -262,7 +287,7 @@ class FakeApiServer:
                 if self._maybe_throttle():
                     return
                 if not self._auth_ok():
-                    return self._send(401, {"kind": "Status", "code": 401})
+                    return self._send(401, self._status(401, "Unauthorized", "Unauthorized"))
                 length = int(self.headers.get("Content-Length", "0"))
                 obj = json.loads(self.rfile.read(length) or b"{}")
                 r = self._route()
@@ -270,7 +295,11 @@ class FakeApiServer:
                     return self._send(404, {"kind": "Status", "code": 404})
                 kind, ns, _name, _sub = r
                 if kind == "Event":
+                    obj.setdefault("metadata", {})
+                    obj["metadata"].setdefault("uid", str(uuid.uuid4()))
+                    obj["metadata"].setdefault("creationTimestamp", _now_rfc3339())
                     with fixture._lock:
+                        obj["metadata"].setdefault("resourceVersion", str(fixture._rv))
                         fixture.events.append(obj)
                     return self._send(201, obj)
                 name = obj.get("metadata", {}).get("name", "")

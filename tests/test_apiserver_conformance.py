@@ -1,0 +1,74 @@
+"""Apiserver conformance: the recorded-transcript suite.
+
+No kube-apiserver binary or network exists in this environment, so the
+hermetic evidence that FakeApiServer matches real apiserver semantics is
+this transcript suite: request/response captures authored from the
+Kubernetes API conventions (metav1.Status shapes, autoscaling/v1 Scale
+subresource, RFC 7386 merge-patch, Event defaulting, List envelopes),
+replayed against the fake. The SAME transcripts replay against a real
+kube-apiserver in tests/test_envtest_e2e.py whenever envtest binaries are
+present — one source of truth for both tiers.
+"""
+
+import json
+
+import pytest
+
+from gpu_pruner_amd.fixtures import FakeApiServer
+from gpu_pruner_amd.fixtures.conformance_replay import replay_step, transcripts
+
+TRANSCRIPTS = transcripts()
+assert TRANSCRIPTS, "no conformance transcripts found"
+
+
+@pytest.mark.parametrize("path", TRANSCRIPTS, ids=lambda p: p.stem)
+def test_fake_apiserver_conformance(path):
+    t = json.loads(path.read_text())
+    fixture_cfg = t.get("fixture", {})
+    with FakeApiServer(token=fixture_cfg.get("token")) as api:
+        for obj in t.get("seed", []):
+            api.put(obj["kind"], json.loads(json.dumps(obj)))
+        default_headers = {}
+        if fixture_cfg.get("token"):
+            default_headers["Authorization"] = f"Bearer {fixture_cfg['token']}"
+        failures = []
+        for i, step in enumerate(t["steps"]):
+            if step["request"].get("inject_throttle"):
+                api.throttle_next = step["request"]["inject_throttle"]
+                api.retry_after_s = 0
+            errs = replay_step(api.url, step, default_headers)
+            failures.extend(f"step {i} ({step['request']['method']} "
+                            f"{step['request']['path']}): {e}" for e in errs)
+        assert not failures, "\n".join(failures)
+
+
+def test_transcripts_cover_every_pruner_verb():
+    """The suite must exercise every apiserver interaction the pruner
+    performs: GET (found + 404), LIST, merge-PATCH, /scale PATCH, Event
+    POST, 401 and 429 — so fake-fidelity claims cover the whole surface."""
+    seen = {"get_200": False, "get_404": False, "list": False,
+            "merge_patch": False, "scale": False, "event_post": False,
+            "unauthorized": False, "throttle": False}
+    for path in TRANSCRIPTS:
+        t = json.loads(path.read_text())
+        for step in t["steps"]:
+            req, exp = step["request"], step["expect"]
+            p, m, s = req["path"], req["method"], exp["status"]
+            if m == "GET" and s == 200 and not p.rstrip("/").endswith(("pods", "events")):
+                seen["get_200"] = True
+            if m == "GET" and s == 404:
+                seen["get_404"] = True
+            if m == "GET" and exp.get("body", {}).get("kind", "").endswith("List"):
+                seen["list"] = True
+            if m == "PATCH" and "/scale" in p and s == 200:
+                seen["scale"] = True
+            if m == "PATCH" and "/scale" not in p and s == 200:
+                seen["merge_patch"] = True
+            if m == "POST" and "/events" in p and s == 201:
+                seen["event_post"] = True
+            if s == 401:
+                seen["unauthorized"] = True
+            if s == 429:
+                seen["throttle"] = True
+    missing = [k for k, v in seen.items() if not v]
+    assert not missing, f"conformance gaps: {missing}"

@@ -1,0 +1,310 @@
+"""Real-apiserver e2e tier (envtest-style), gated on binaries being present.
+
+The reference validates its semantics against a live kind cluster
+(gpu-pruner/tests/e2e.rs, ci.yml:49-55). This environment has no network to
+fetch kube binaries, so:
+
+  * the transcript suite (test_apiserver_conformance.py) always runs against
+    the fake, and
+  * THIS module lights up when envtest assets exist — set
+    ``KUBEBUILDER_ASSETS`` to a directory containing ``etcd`` and
+    ``kube-apiserver`` (what ``setup-envtest use -p path`` produces) — and
+    then (a) replays the SAME conformance transcripts against the real
+    apiserver and (b) ports the reference's e2e scenarios: owner-walk
+    resolution, scale-to-zero on Deployment/StatefulSet, Notebook stop
+    annotation and InferenceService minReplicas via the shipped binary, and
+    gpuscaler- Event creation.
+
+Steps marked ``skip_on_real`` in a transcript cover fake-only injection
+hooks (throttle) or fields the real server overrides (seeded uids/status).
+"""
+
+import json
+import os
+import shutil
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+from gpu_pruner_amd.fixtures.conformance_replay import (http_request, replay_step,
+                                                        transcripts)
+
+REPO_ROOT = Path(__file__).resolve().parent.parent
+
+ASSETS = os.environ.get("KUBEBUILDER_ASSETS", "")
+
+
+def _find(binary):
+    if ASSETS and (Path(ASSETS) / binary).exists():
+        return str(Path(ASSETS) / binary)
+    return shutil.which(binary)
+
+
+ETCD = _find("etcd")
+APISERVER = _find("kube-apiserver")
+
+pytestmark = pytest.mark.skipif(
+    not (ETCD and APISERVER),
+    reason="envtest binaries not present (set KUBEBUILDER_ASSETS to a dir "
+           "with etcd + kube-apiserver, e.g. from `setup-envtest use -p path`)")
+
+TOKEN = "envtest-e2e-token"
+
+
+@pytest.fixture(scope="module")
+def real_apiserver(tmp_path_factory):
+    """etcd + kube-apiserver with static token auth, AlwaysAllow authz."""
+    d = tmp_path_factory.mktemp("envtest")
+    procs = []
+
+    def run(cmd):
+        p = subprocess.Popen(cmd, stdout=subprocess.DEVNULL,
+                             stderr=open(d / (Path(cmd[0]).name + ".log"), "w"))
+        procs.append(p)
+        return p
+
+    import socket
+
+    def free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        return port
+
+    etcd_client = free_port()
+    etcd_peer = free_port()
+    api_port = free_port()
+
+    run([ETCD, "--data-dir", str(d / "etcd"),
+         "--listen-client-urls", f"http://127.0.0.1:{etcd_client}",
+         "--advertise-client-urls", f"http://127.0.0.1:{etcd_client}",
+         "--listen-peer-urls", f"http://127.0.0.1:{etcd_peer}"])
+
+    (d / "tokens.csv").write_text(f"{TOKEN},envtest-user,envtest-uid,system:masters\n")
+    # service-account signing key (required by modern apiservers)
+    sa_key = d / "sa.key"
+    subprocess.run(["openssl", "genrsa", "-out", str(sa_key), "2048"],
+                   check=True, capture_output=True)
+
+    run([APISERVER,
+         "--etcd-servers", f"http://127.0.0.1:{etcd_client}",
+         "--secure-port", str(api_port),
+         "--bind-address", "127.0.0.1",
+         "--cert-dir", str(d / "certs"),
+         "--token-auth-file", str(d / "tokens.csv"),
+         "--authorization-mode", "AlwaysAllow",
+         "--service-account-issuer", "https://envtest.local",
+         "--service-account-key-file", str(sa_key),
+         "--service-account-signing-key-file", str(sa_key),
+         "--disable-admission-plugins", "ServiceAccount",
+         "--allow-privileged"])
+
+    base = f"https://127.0.0.1:{api_port}"
+    headers = {"Authorization": f"Bearer {TOKEN}"}
+    deadline = time.monotonic() + 60
+    ready = False
+    while time.monotonic() < deadline:
+        try:
+            status, _, _ = http_request(base, "GET", "/readyz", headers=headers,
+                                        insecure=True)
+            if status == 200:
+                ready = True
+                break
+        except OSError:
+            pass
+        time.sleep(0.5)
+    if not ready:
+        for p in procs:
+            p.terminate()
+        pytest.fail("kube-apiserver did not become ready; see logs in " + str(d))
+
+    # namespaces + CRDs the scenarios need
+    for ns in ("ml", "default"):
+        http_request(base, "POST", "/api/v1/namespaces",
+                     {"apiVersion": "v1", "kind": "Namespace",
+                      "metadata": {"name": ns}}, headers, insecure=True)
+    import yaml
+    for crd_file in sorted((REPO_ROOT / "deploy" / "crds").glob("*.yaml")):
+        crd = yaml.safe_load(crd_file.read_text())
+        http_request(base, "POST",
+                     "/apis/apiextensions.k8s.io/v1/customresourcedefinitions",
+                     crd, headers, insecure=True)
+    time.sleep(2)  # CRD establishment
+
+    yield {"url": base, "headers": headers, "dir": d}
+    for p in procs:
+        p.terminate()
+    for p in procs:
+        try:
+            p.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            p.kill()
+
+
+def _collection_path(obj):
+    api_version = obj["apiVersion"]
+    plural = obj["kind"].lower() + "s"
+    ns = obj["metadata"]["namespace"]
+    if "/" in api_version:
+        return f"/apis/{api_version}/namespaces/{ns}/{plural}"
+    return f"/api/{api_version}/namespaces/{ns}/{plural}"
+
+
+@pytest.mark.parametrize("path", transcripts(), ids=lambda p: p.stem)
+def test_transcripts_against_real_apiserver(real_apiserver, path):
+    t = json.loads(path.read_text())
+    base, headers = real_apiserver["url"], real_apiserver["headers"]
+    for obj in t.get("seed", []):
+        obj = json.loads(json.dumps(obj))
+        obj["metadata"].pop("uid", None)  # server-assigned
+        obj["metadata"].pop("resourceVersion", None)
+        status, body, _ = http_request(base, "POST", _collection_path(obj), obj,
+                                       headers, insecure=True)
+        assert status in (200, 201, 409), (status, body)
+    failures = []
+    for i, step in enumerate(t["steps"]):
+        if step.get("skip_on_real") or step["request"].get("inject_throttle"):
+            continue
+        errs = replay_step(base, step, headers, insecure=True)
+        failures.extend(f"step {i}: {e}" for e in errs)
+    assert not failures, "\n".join(failures)
+
+
+def _run_pruner(real_apiserver, fake_prom, *args):
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = real_apiserver["url"]
+    env["GPU_PRUNER_K8S_TOKEN"] = TOKEN
+    env["GPU_PRUNER_K8S_SKIP_TLS"] = "1"
+    env["PROMETHEUS_TOKEN"] = "t"
+    return subprocess.run(
+        [str(REPO_ROOT / "bin" / "gpu-pruner"), "--prometheus-url", fake_prom.url,
+         "--run-mode", "scale-down", *args],
+        capture_output=True, text=True, timeout=120, env=env)
+
+
+def _seed_workload(real_apiserver, kind_objs):
+    base, headers = real_apiserver["url"], real_apiserver["headers"]
+    created = []
+    for obj in kind_objs:
+        status, body, _ = http_request(base, "POST", _collection_path(obj), obj,
+                                       headers, insecure=True)
+        assert status in (200, 201), (status, body)
+        created.append(body)
+    return created
+
+
+def test_owner_walk_scales_deployment_not_replicaset(real_apiserver, fake_prom):
+    """Reference e2e.rs:168-197 + 256-333: the walk resolves the Deployment
+    root and scales IT to zero; a gpuscaler- Event lands in the cluster."""
+    base, headers = real_apiserver["url"], real_apiserver["headers"]
+    dep, = _seed_workload(real_apiserver, [{
+        "apiVersion": "apps/v1", "kind": "Deployment",
+        "metadata": {"name": "e2e-dep", "namespace": "ml"},
+        "spec": {"replicas": 1,
+                 "selector": {"matchLabels": {"app": "e2e"}},
+                 "template": {"metadata": {"labels": {"app": "e2e"}},
+                               "spec": {"containers": [
+                                   {"name": "c", "image": "pause:3.10"}]}}},
+    }])
+    rs, = _seed_workload(real_apiserver, [{
+        "apiVersion": "apps/v1", "kind": "ReplicaSet",
+        "metadata": {"name": "e2e-dep-rs", "namespace": "ml",
+                     "ownerReferences": [{
+                         "apiVersion": "apps/v1", "kind": "Deployment",
+                         "name": "e2e-dep", "uid": dep["metadata"]["uid"]}]},
+        "spec": {"replicas": 1,
+                 "selector": {"matchLabels": {"app": "e2e"}},
+                 "template": {"metadata": {"labels": {"app": "e2e"}},
+                               "spec": {"containers": [
+                                   {"name": "c", "image": "pause:3.10"}]}}},
+    }])
+    _seed_workload(real_apiserver, [{
+        "apiVersion": "v1", "kind": "Pod",
+        "metadata": {"name": "e2e-pod-0", "namespace": "ml",
+                     "ownerReferences": [{
+                         "apiVersion": "apps/v1", "kind": "ReplicaSet",
+                         "name": "e2e-dep-rs", "uid": rs["metadata"]["uid"]}]},
+        "spec": {"containers": [{"name": "c", "image": "pause:3.10"}]},
+    }])
+    # eligibility needs pod age >= duration + grace; the real server stamps
+    # creationTimestamp "now", so wait out a 1-minute window once
+    fake_prom.add_idle_series("e2e-pod-0", "ml")
+    time.sleep(65)
+    r = _run_pruner(real_apiserver, fake_prom, "-t", "1", "--grace-period", "0")
+    assert r.returncode == 0, r.stderr
+    status, dep_after, _ = http_request(base, "GET",
+                                        "/apis/apps/v1/namespaces/ml/deployments/e2e-dep",
+                                        headers=headers, insecure=True)
+    assert status == 200
+    assert dep_after["spec"]["replicas"] == 0
+    status, rs_after, _ = http_request(base, "GET",
+                                       "/apis/apps/v1/namespaces/ml/replicasets/e2e-dep-rs",
+                                       headers=headers, insecure=True)
+    assert rs_after["spec"]["replicas"] == 1  # the RS itself is untouched
+    status, events, _ = http_request(base, "GET", "/api/v1/namespaces/ml/events",
+                                     headers=headers, insecure=True)
+    names = [e["metadata"]["name"] for e in events.get("items", [])]
+    assert any(n.startswith("gpuscaler-") for n in names), names
+
+
+def test_notebook_and_inferenceservice_crd_paths(real_apiserver, fake_prom):
+    """Reference gap closed (kind e2e never installed the CRDs): Notebook
+    stop-annotation and InferenceService minReplicas patches against a real
+    apiserver with the CRDs established."""
+    base, headers = real_apiserver["url"], real_apiserver["headers"]
+    nb, = _seed_workload(real_apiserver, [{
+        "apiVersion": "kubeflow.org/v1", "kind": "Notebook",
+        "metadata": {"name": "e2e-wb", "namespace": "ml"},
+        "spec": {"template": {"spec": {"containers": []}}},
+    }])
+    _seed_workload(real_apiserver, [{
+        "apiVersion": "apps/v1", "kind": "StatefulSet",
+        "metadata": {"name": "e2e-wb", "namespace": "ml",
+                     "ownerReferences": [{
+                         "apiVersion": "kubeflow.org/v1", "kind": "Notebook",
+                         "name": "e2e-wb", "uid": nb["metadata"]["uid"]}]},
+        "spec": {"replicas": 1, "serviceName": "e2e-wb",
+                 "selector": {"matchLabels": {"app": "wb"}},
+                 "template": {"metadata": {"labels": {"app": "wb"}},
+                               "spec": {"containers": [
+                                   {"name": "c", "image": "pause:3.10"}]}}},
+    }])
+    _seed_workload(real_apiserver, [{
+        "apiVersion": "serving.kserve.io/v1beta1", "kind": "InferenceService",
+        "metadata": {"name": "e2e-svc", "namespace": "ml",
+                     "labels": {"serving.kserve.io/inferenceservice": "e2e-svc"}},
+        "spec": {"predictor": {"minReplicas": 1}},
+    }])
+    status, ss_obj, _ = http_request(base, "GET",
+                                     "/apis/apps/v1/namespaces/ml/statefulsets/e2e-wb",
+                                     headers=headers, insecure=True)
+    _seed_workload(real_apiserver, [{
+        "apiVersion": "v1", "kind": "Pod",
+        "metadata": {"name": "e2e-wb-0", "namespace": "ml",
+                     "ownerReferences": [{
+                         "apiVersion": "apps/v1", "kind": "StatefulSet",
+                         "name": "e2e-wb", "uid": ss_obj["metadata"]["uid"]}]},
+        "spec": {"containers": [{"name": "c", "image": "pause:3.10"}]},
+    }, {
+        "apiVersion": "v1", "kind": "Pod",
+        "metadata": {"name": "e2e-svc-0", "namespace": "ml",
+                     "labels": {"serving.kserve.io/inferenceservice": "e2e-svc"}},
+        "spec": {"containers": [{"name": "c", "image": "pause:3.10"}]},
+    }])
+    fake_prom.add_idle_series("e2e-wb-0", "ml")
+    fake_prom.add_idle_series("e2e-svc-0", "ml")
+    time.sleep(65)
+    r = _run_pruner(real_apiserver, fake_prom, "-t", "1", "--grace-period", "0")
+    assert r.returncode == 0, r.stderr
+    status, nb_after, _ = http_request(
+        base, "GET", "/apis/kubeflow.org/v1/namespaces/ml/notebooks/e2e-wb",
+        headers=headers, insecure=True)
+    assert "kubeflow-resource-stopped" in nb_after["metadata"].get("annotations", {})
+    status, svc_after, _ = http_request(
+        base, "GET",
+        "/apis/serving.kserve.io/v1beta1/namespaces/ml/inferenceservices/e2e-svc",
+        headers=headers, insecure=True)
+    assert svc_after["spec"]["predictor"]["minReplicas"] == 0
