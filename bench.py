@@ -64,6 +64,9 @@ def main():
                     help="export spans/metrics to an in-process OTLP collector "
                          "during the timed region (BASELINE config 5; on by "
                          "default, --no-otlp for the bare-engine figure)")
+    ap.add_argument("--strategy", default="auto",
+                    choices=["get", "list", "auto", "watch"],
+                    help="engine --eval-strategy for the timed region")
     ap.add_argument("--rtt-point-us", type=int, default=2000,
                     help="after the primary (loopback, CPU-bound) measurement, "
                          "also measure a few ticks at this injected apiserver "
@@ -127,7 +130,7 @@ def main():
     cfg = json.dumps({
         "duration": 30, "grace_period": 300, "run_mode": "scale-down",
         "prometheus_url": backend.prom_url, "max_concurrency": args.concurrency,
-        "model_name": "AMD Instinct MI355X",
+        "model_name": "AMD Instinct MI355X", "eval_strategy": args.strategy,
     })
 
     # ---- real-GPU utilization feed (the rank's own device) ----
@@ -266,6 +269,7 @@ def main():
                 "n_pods": pods_per_rank * n_gpus,
                 "pods_per_rank": pods_per_rank,
                 "parallelism": f"rank-sharded x{n_gpus}",
+                "eval_strategy": args.strategy,
                 "max_concurrency": args.concurrency,
                 "apiserver_latency_us": args.latency_us,
                 # full-tick latency (query -> eval -> walks -> actuation for

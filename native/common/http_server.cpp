@@ -9,6 +9,7 @@
 
 #include <chrono>
 #include <condition_variable>
+#include <cstdio>
 #include <cstring>
 #include <mutex>
 #include <set>
@@ -182,19 +183,52 @@ void Server::handle_conn(int fd) {
                          : resp.status == 404 ? "Not Found"
                          : resp.status == 400 ? "Bad Request"
                                               : "Status";
+    auto send_all = [&](const std::string& data) {
+      size_t off = 0;
+      while (off < data.size()) {
+        ssize_t w = ::send(fd, data.data() + off, data.size() - off, MSG_NOSIGNAL);
+        if (w <= 0) return false;
+        off += static_cast<size_t>(w);
+      }
+      return true;
+    };
+
+    if (resp.streamer) {
+      // chunked streaming response (watch): one chunk per streamer call
+      std::string head = "HTTP/1.1 " + std::to_string(resp.status) + " " + reason + "\r\n";
+      head += "Content-Type: " + resp.content_type + "\r\n";
+      head += "Transfer-Encoding: chunked\r\n";
+      head += "Connection: close\r\n\r\n";
+      if (!send_all(head)) {
+        ::close(fd);
+        return;
+      }
+      bool more = true;
+      while (more && running_.load()) {
+        std::string chunk;
+        more = resp.streamer(&chunk);
+        if (!chunk.empty()) {
+          char sz[16];
+          std::snprintf(sz, sizeof sz, "%zx\r\n", chunk.size());
+          if (!send_all(std::string(sz) + chunk + "\r\n")) {
+            ::close(fd);
+            return;
+          }
+        }
+      }
+      send_all("0\r\n\r\n");
+      ::close(fd);
+      return;
+    }
+
     std::string out = "HTTP/1.1 " + std::to_string(resp.status) + " " + reason + "\r\n";
     out += "Content-Type: " + resp.content_type + "\r\n";
     out += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
     out += "Connection: keep-alive\r\n\r\n";
     out += resp.body;
-    size_t off = 0;
-    while (off < out.size()) {
-      ssize_t w = ::send(fd, out.data() + off, out.size() - off, MSG_NOSIGNAL);
-      if (w <= 0) {
-        ::close(fd);
-        return;
-      }
-      off += static_cast<size_t>(w);
+    if (!send_all(out)) {
+      ::close(fd);
+      return;
     }
   }
   ::close(fd);

@@ -30,6 +30,14 @@ struct ServerResponse {
   int status = 200;
   std::string content_type = "text/plain; charset=utf-8";
   std::string body;
+  // Streaming mode (Kubernetes watch): when set, `body` is ignored and the
+  // response is sent with chunked transfer encoding — the function is
+  // called repeatedly from the connection thread; each non-empty *chunk is
+  // written as one chunk; returning false ends the stream (after writing
+  // any final chunk). The connection closes after a streamed response.
+  // Implementations must bound their internal waits so server stop() can
+  // drain (they are invoked on detached connection threads).
+  std::function<bool(std::string* chunk)> streamer;
 };
 
 class Server {

@@ -5,6 +5,7 @@
 #include <condition_variable>
 #include <mutex>
 #include <memory>
+#include <random>
 #include <thread>
 #include <vector>
 
@@ -73,9 +74,26 @@ uint64_t now_unix_ns() {
                                    .count());
 }
 
+// Span/trace ids need uniqueness, not cryptographic strength; the hot path
+// mints ~3k ids per 1000-pod tick, so avoid uuid4_simple's per-call
+// /dev/urandom open/read/close — a thread_local PRNG seeded once from the
+// OS is plenty.
 std::string rand_hex(size_t bytes) {
-  std::string id = strutil::uuid4_simple();  // 32 hex chars
-  return id.substr(0, bytes * 2);
+  static thread_local std::mt19937_64 rng{[] {
+    std::random_device rd;
+    return (static_cast<uint64_t>(rd()) << 32) ^ rd() ^
+           std::hash<std::thread::id>{}(std::this_thread::get_id());
+  }()};
+  static const char* hex = "0123456789abcdef";
+  std::string out(bytes * 2, '0');
+  for (size_t i = 0; i < bytes; i += 8) {
+    uint64_t v = rng();
+    for (size_t b = 0; b < 8 && i + b < bytes; b++) {
+      out[(i + b) * 2] = hex[(v >> (b * 8 + 4)) & 0xF];
+      out[(i + b) * 2 + 1] = hex[(v >> (b * 8)) & 0xF];
+    }
+  }
+  return out;
 }
 
 std::string hex_to_bytes(const std::string& hex) {

@@ -2,6 +2,8 @@
 
 #include <time.h>
 
+#include <algorithm>
+#include <chrono>
 #include <cstdio>
 
 #include "../common/strutil.hpp"
@@ -160,6 +162,11 @@ void SyntheticBackend::start() {
 }
 
 void SyntheticBackend::stop() {
+  closing_.store(true);
+  {
+    std::lock_guard<std::mutex> lock(mu_);  // pair with streamer waits
+  }
+  event_cv_.notify_all();
   if (prom_server_) prom_server_->stop();
   if (k8s_server_) k8s_server_->stop();
 }
@@ -255,11 +262,56 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
     return resp;
   }
 
+  // Kubernetes watch: chunked stream of events after ?resourceVersion,
+  // served by the generic streamer path (informer / --eval-strategy watch).
+  if (req.method == "GET" && name.empty() &&
+      req.query.find("watch=true") != std::string::npos) {
+    uint64_t since = 0;
+    if (size_t p = req.query.find("resourceVersion="); p != std::string::npos)
+      since = std::strtoull(req.query.c_str() + p + 16, nullptr, 10);
+    double timeout_s = 30.0;
+    if (size_t p = req.query.find("timeoutSeconds="); p != std::string::npos)
+      timeout_s = std::strtod(req.query.c_str() + p + 15, nullptr);
+    auto deadline = std::chrono::steady_clock::now() +
+                    std::chrono::duration_cast<std::chrono::steady_clock::duration>(
+                        std::chrono::duration<double>(timeout_s));
+    watch_streams_.fetch_add(1, std::memory_order_relaxed);
+    bool bookmark_sent = false;
+    resp.streamer = [this, kind, ns, since, deadline,
+                     bookmark_sent](std::string* chunk) mutable {
+      std::unique_lock<std::mutex> lock(mu_);
+      while (true) {
+        // log rvs are strictly increasing: binary-search the resume point
+        auto it = std::lower_bound(watch_log_.begin(), watch_log_.end(), since + 1,
+                                   [](const WatchEvent& e, uint64_t rv) { return e.rv < rv; });
+        for (; it != watch_log_.end(); ++it) {
+          if (it->kind == kind && it->ns == ns) {
+            since = it->rv;
+            *chunk = it->line + "\n";
+            return true;
+          }
+          since = it->rv;
+        }
+        if (bookmark_sent) return false;
+        if (closing_.load() || std::chrono::steady_clock::now() >= deadline) {
+          *chunk = "{\"type\":\"BOOKMARK\",\"object\":{\"kind\":\"" + kind +
+                   "\",\"metadata\":{\"resourceVersion\":\"" + std::to_string(rv_) +
+                   "\"}}}\n";
+          bookmark_sent = true;
+          return true;  // deliver the bookmark; next call ends the stream
+        }
+        event_cv_.wait_for(lock, std::chrono::milliseconds(100));
+      }
+    };
+    return resp;
+  }
+
   std::lock_guard<std::mutex> lock(mu_);
   auto kit = objects_.find(kind);
   // namespaced collection LIST (no object name)
   if (req.method == "GET" && name.empty()) {
-    std::string body = "{\"kind\":\"" + kind + "List\",\"items\":[";
+    std::string body = "{\"kind\":\"" + kind + "List\",\"metadata\":{\"resourceVersion\":\"" +
+                       std::to_string(rv_) + "\"},\"items\":[";
     bool first = true;
     if (kit != objects_.end()) {
       auto nit = kit->second.find(ns);
@@ -296,16 +348,26 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
   }
   if (req.method == "PATCH") {
     jsn::Value patch = jsn::parse(req.body);
+    auto record_modified = [&] {
+      rv_++;
+      obj->obj["metadata"]["resourceVersion"] = std::to_string(rv_);
+      obj->cached_dump.clear();
+      obj->cached_dump = obj->obj.dump();
+      watch_log_.push_back(
+          {rv_, kind, ns, "{\"type\":\"MODIFIED\",\"object\":" + obj->cached_dump + "}"});
+      while (watch_log_.size() > 100000) watch_log_.pop_front();
+      event_cv_.notify_all();
+    };
     if (is_scale) {
       scale_patches_.fetch_add(1, std::memory_order_relaxed);
       obj->obj["spec"]["replicas"] = patch.at({"spec", "replicas"});
-      obj->cached_dump.clear();
+      record_modified();
       resp.body = "{\"kind\":\"Scale\",\"spec\":" + patch.get("spec").dump() + "}";
       return resp;
     }
     scale_patches_.fetch_add(1, std::memory_order_relaxed);
     obj->obj.merge_patch(patch);
-    obj->cached_dump.clear();
+    record_modified();
     resp.body = obj->obj.dump();
     return resp;
   }

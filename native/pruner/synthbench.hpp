@@ -11,6 +11,8 @@
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
+#include <deque>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -47,6 +49,7 @@ public:
   int64_t events_posted() const { return events_posted_.load(); }
   int64_t scale_patches() const { return scale_patches_.load(); }
   int64_t requests_served() const { return requests_.load(); }
+  int64_t watch_streams() const { return watch_streams_.load(); }
   int expected_parents() const { return n_parents_; }
 
 private:
@@ -64,6 +67,17 @@ private:
   // kind → ns → name → object
   std::map<std::string, std::map<std::string, std::map<std::string, StoredObject>>> objects_;
   std::string series_json_zero_;  // pre-rendered result vector (values patched in)
+  // watch support: monotonic resourceVersion + event log (pre-serialized
+  // lines), mirroring the Python fixture's semantics
+  std::condition_variable event_cv_;
+  uint64_t rv_ = 1;
+  struct WatchEvent {
+    uint64_t rv;
+    std::string kind, ns, line;
+  };
+  std::deque<WatchEvent> watch_log_;
+  std::atomic<bool> closing_{false};
+  std::atomic<int64_t> watch_streams_{0};
   std::atomic<double> series_value_{0.0};
   std::atomic<int64_t> events_posted_{0};
   std::atomic<int64_t> scale_patches_{0};

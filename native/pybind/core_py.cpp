@@ -309,6 +309,7 @@ PYBIND11_MODULE(_pruner_core, m) {
       .def("set_series_value", &SyntheticBackend::set_series_value)
       .def_property_readonly("events_posted", &SyntheticBackend::events_posted)
       .def_property_readonly("scale_patches", &SyntheticBackend::scale_patches)
+      .def_property_readonly("watch_streams", &SyntheticBackend::watch_streams)
       .def_property_readonly("requests_served", &SyntheticBackend::requests_served)
       .def_property_readonly("expected_parents", &SyntheticBackend::expected_parents);
 
