@@ -341,7 +341,7 @@ std::optional<jsn::Value> KubeClient::get_object(Kind kind, const std::string& n
   return get_opt(object_path(kind, ns, name));
 }
 
-jsn::Value KubeClient::merge_patch(const std::string& path, const jsn::Value& patch) {
+void KubeClient::merge_patch(const std::string& path, const jsn::Value& patch) {
   http::Request r;
   r.method = "PATCH";
   r.path = path;
@@ -351,12 +351,11 @@ jsn::Value KubeClient::merge_patch(const std::string& path, const jsn::Value& pa
   if (resp.status < 200 || resp.status >= 300)
     throw KubeError(resp.status, "PATCH " + path + " -> " + std::to_string(resp.status) + ": " +
                                      resp.body.substr(0, 300));
-  return jsn::parse(resp.body);
 }
 
-jsn::Value KubeClient::patch_scale(Kind kind, const std::string& ns, const std::string& name,
-                                   const jsn::Value& patch) {
-  return merge_patch(object_path(kind, ns, name) + "/scale", patch);
+void KubeClient::patch_scale(Kind kind, const std::string& ns, const std::string& name,
+                             const jsn::Value& patch) {
+  merge_patch(object_path(kind, ns, name) + "/scale", patch);
 }
 
 std::unique_ptr<http::BodyStream> KubeClient::open_stream(const std::string& path) {
@@ -368,7 +367,7 @@ std::unique_ptr<http::BodyStream> KubeClient::open_stream(const std::string& pat
   return http_->open_stream(r);
 }
 
-jsn::Value KubeClient::create(const std::string& collection, const jsn::Value& obj) {
+void KubeClient::create(const std::string& collection, const jsn::Value& obj) {
   http::Request r;
   r.method = "POST";
   r.path = collection;
@@ -378,7 +377,6 @@ jsn::Value KubeClient::create(const std::string& collection, const jsn::Value& o
   if (resp.status < 200 || resp.status >= 300)
     throw KubeError(resp.status, "POST " + collection + " -> " + std::to_string(resp.status) +
                                      ": " + resp.body.substr(0, 300));
-  return jsn::parse(resp.body);
 }
 
 }  // namespace pruner
