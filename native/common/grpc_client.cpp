@@ -208,6 +208,10 @@ void hpack_literal_new_name(std::string& out, const std::string& name,
 
 std::string unary_call(const Target& target, const std::string& method_path,
                        const std::string& request_msg, int timeout_ms) {
+  if (request_msg.size() > kMaxRequestBytes)
+    throw GrpcError("request of " + std::to_string(request_msg.size()) +
+                    " bytes exceeds the flow-control-safe limit (" +
+                    std::to_string(kMaxRequestBytes) + "); split into multiple calls");
   Sock sock(target, timeout_ms);
 
   // ---- connection preface + SETTINGS + generous connection window ----

@@ -28,11 +28,19 @@ struct Target {
   std::string authority = "localhost";  // :authority pseudo-header
 };
 
+// Largest request message this client will send. The client writes the whole
+// request before reading any frames, so it must stay inside the peer's
+// DEFAULT stream flow-control window (65535 bytes, RFC 9113 §6.9.2) — we
+// never see the server's WINDOW_UPDATEs in time to send more. Callers with
+// bigger payloads must split them into multiple unary calls.
+constexpr size_t kMaxRequestBytes = 60000;
+
 // One unary call: sends `request_msg` (raw protobuf message bytes, framing
 // added here) to `method_path` (e.g.
 // "/opentelemetry.proto.collector.trace.v1.TraceService/Export") and returns
 // the concatenated response message bytes (gRPC frames unwrapped). Throws
-// GrpcError on connect/transport errors, stream reset, or deadline.
+// GrpcError on connect/transport errors, stream reset, deadline, or a
+// request larger than kMaxRequestBytes.
 std::string unary_call(const Target& target, const std::string& method_path,
                        const std::string& request_msg, int timeout_ms);
 

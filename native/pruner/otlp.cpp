@@ -252,10 +252,24 @@ void export_once() {
     std::vector<FinishedSpan> batch(all.begin() + static_cast<long>(base),
                                     all.begin() + static_cast<long>(base + n));
     if (s.transport == Transport::Grpc) {
-      try {
-        grpc_export(s, "/opentelemetry.proto.collector.trace.v1.TraceService/Export",
-                    encode_spans_pb(s, batch));
-      } catch (const std::exception&) { /* collector away; drop batch */ }
+      // size-bounded sub-batches: each Export request must fit the peer's
+      // default HTTP/2 flow-control window (grpcx::kMaxRequestBytes)
+      size_t start = 0, take = batch.size();
+      while (start < batch.size()) {
+        take = std::min(take, batch.size() - start);
+        std::vector<FinishedSpan> sub(batch.begin() + static_cast<long>(start),
+                                      batch.begin() + static_cast<long>(start + take));
+        std::string msg = encode_spans_pb(s, sub);
+        if (msg.size() > grpcx::kMaxRequestBytes && take > 1) {
+          take /= 2;
+          continue;
+        }
+        try {
+          grpc_export(s, "/opentelemetry.proto.collector.trace.v1.TraceService/Export",
+                      std::move(msg));
+        } catch (const std::exception&) { /* collector away; drop batch */ }
+        start += take;
+      }
       continue;
     }
     if (s.transport == Transport::HttpProtobuf) {

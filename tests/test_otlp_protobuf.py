@@ -6,6 +6,7 @@ opentelemetry-proto field numbers — an independent implementation of the
 wire format, mirroring the PodResources decoder test in reverse.
 """
 
+import json
 import os
 import subprocess
 
@@ -247,3 +248,43 @@ def test_json_transport_also_carries_parent_ids(pruner_bin, fake_api, fake_prom)
         for s in spans:
             if s["name"] == "scale_to_zero":
                 assert by_id[s["parentSpanId"]]["name"] == "scale"
+
+
+def test_grpc_large_batch_respects_flow_control(core, fake_api, monkeypatch):
+    """A 1000-pod tick mints thousands of spans; the gRPC exporter must split
+    Export requests to fit the peer's default HTTP/2 flow-control window
+    (65535 B) — a single oversized write would be a protocol violation the
+    collector tears down."""
+    from gpu_pruner_amd.fixtures import FakeOtlpGrpcCollector
+
+    with FakeOtlpGrpcCollector() as collector:
+        monkeypatch.setenv("OTEL_EXPORTER_OTLP_ENDPOINT", collector.url)
+        monkeypatch.setenv("OTEL_EXPORTER_OTLP_PROTOCOL", "grpc")
+        monkeypatch.setenv("OTEL_METRIC_EXPORT_INTERVAL", "60000")
+        monkeypatch.setenv("PROMETHEUS_TOKEN", "t")
+        b = core.SyntheticBackend(n_pods=1000, pods_per_parent=2)
+        b.start()
+        try:
+            monkeypatch.setenv("GPU_PRUNER_K8S_URL", b.k8s_url)
+            core.otlp_init("flowcontrol-test")
+            cfg = json.dumps({"duration": 30, "grace_period": 300,
+                              "run_mode": "scale-down",
+                              "prometheus_url": b.prom_url})
+            out = core.run_tick(cfg)
+            assert out["num_unique_pods"] == 1000
+            core.otlp_shutdown()
+        finally:
+            b.stop()
+        assert len(collector.traces_pb) >= 2, "large batch was not split"
+        for raw in collector.traces_pb:
+            assert len(raw) <= 60000, f"oversized Export request: {len(raw)} B"
+        TracesData, _ = _build_otlp_messages()
+        total = 0
+        for raw in collector.traces_pb:
+            td = TracesData()
+            td.ParseFromString(raw)
+            for rsp in td.resource_spans:
+                for ss in rsp.scope_spans:
+                    total += len(ss.spans)
+        # 1000 find_root_object + 500 x (scale + helper + event) + roots
+        assert total >= 2500, total
