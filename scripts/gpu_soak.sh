@@ -19,8 +19,10 @@ backend = core.SyntheticBackend(n_pods=100)
 backend.start()
 os.environ["GPU_PRUNER_K8S_URL"] = backend.k8s_url
 
+# short sliding window so busy bursts age out inside the phase cadence
 exporter = subprocess.Popen(["./bin/mi355-exporter", "-p", "19420", "-b", "127.0.0.1",
-                             "-i", "250"], stdout=subprocess.DEVNULL,
+                             "-i", "250", "--activity-window", "3"],
+                            stdout=subprocess.DEVNULL,
                             stderr=subprocess.DEVNULL)
 time.sleep(2)
 
@@ -35,8 +37,22 @@ def scrape_ratio():
             return float(line.rsplit("} ", 1)[1])
     raise AssertionError("no series")
 
+# round 2: production daemon shape — watch informers + OTLP export on
+from gpu_pruner_amd.fixtures import FakeOtlpCollector
+collector = FakeOtlpCollector().start()
+os.environ["OTEL_EXPORTER_OTLP_ENDPOINT"] = collector.url
+os.environ["OTEL_METRIC_EXPORT_INTERVAL"] = "2000"
+core.otlp_init("gpu-pruner-soak")
 cfg = json.dumps({"duration": 30, "grace_period": 300, "run_mode": "scale-down",
-                  "prometheus_url": backend.prom_url})
+                  "prometheus_url": backend.prom_url, "eval_strategy": "watch"})
+
+def rss_fds():
+    rss = 0
+    with open("/proc/self/status") as f:
+        for line in f:
+            if line.startswith("VmRSS:"):
+                rss = int(line.split()[1])
+    return rss, len(os.listdir("/proc/self/fd"))
 
 phases = []   # (phase, ticks, scaled_total)
 import os as _os
@@ -66,8 +82,9 @@ try:
                 if scrape_ratio() == 0.0:
                     break
         phases.append(("busy" if busy_phase else "idle", ticks, scaled))
+        rss, fds = rss_fds()
         print(f"phase {phase_idx} ({'busy' if busy_phase else 'idle'}): "
-              f"{ticks} ticks, {scaled} scale actions", flush=True)
+              f"{ticks} ticks, {scaled} scale actions, rss={rss} KiB fds={fds}", flush=True)
         phase_idx += 1
 finally:
     try:
@@ -75,7 +92,12 @@ finally:
     except Exception:
         pass
     exporter.terminate(); exporter.wait()
+    core.otlp_shutdown()
+    spans = len(collector.span_names())
+    collector.stop()
+    core.informers_reset()
     sampler.stop(); backend.stop()
+    print(f"otlp spans exported during soak: {spans}", flush=True)
 
 for phase, ticks, scaled in phases:
     if phase == "busy" and scaled != 0:
