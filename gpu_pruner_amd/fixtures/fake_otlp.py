@@ -82,6 +82,19 @@ class FakeOtlpCollector:
         self.stop()
 
     # -- helpers --------------------------------------------------------------
+    def drain(self) -> tuple[int, int]:
+        """Drop recorded payloads (long soaks would otherwise measure the
+        fixture's own accumulation as a 'leak'); returns (n_traces, n_metrics)
+        dropped."""
+        with self._lock:
+            n = (len(self.traces) + len(self.traces_pb),
+                 len(self.metrics) + len(self.metrics_pb))
+            self.traces.clear()
+            self.metrics.clear()
+            self.traces_pb.clear()
+            self.metrics_pb.clear()
+            return n
+
     def span_names(self) -> list[str]:
         with self._lock:
             names = []

@@ -355,7 +355,7 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
       obj->cached_dump = obj->obj.dump();
       watch_log_.push_back(
           {rv_, kind, ns, "{\"type\":\"MODIFIED\",\"object\":" + obj->cached_dump + "}"});
-      while (watch_log_.size() > 100000) watch_log_.pop_front();
+      while (watch_log_.size() > 20000) watch_log_.pop_front();
       event_cv_.notify_all();
     };
     if (is_scale) {

@@ -43,6 +43,7 @@ collector = FakeOtlpCollector().start()
 os.environ["OTEL_EXPORTER_OTLP_ENDPOINT"] = collector.url
 os.environ["OTEL_METRIC_EXPORT_INTERVAL"] = "2000"
 core.otlp_init("gpu-pruner-soak")
+spans_total = [0]
 cfg = json.dumps({"duration": 30, "grace_period": 300, "run_mode": "scale-down",
                   "prometheus_url": backend.prom_url, "eval_strategy": "watch"})
 
@@ -82,6 +83,7 @@ try:
                 if scrape_ratio() == 0.0:
                     break
         phases.append(("busy" if busy_phase else "idle", ticks, scaled))
+        spans_total[0] += collector.drain()[0]  # don't measure fixture growth
         rss, fds = rss_fds()
         print(f"phase {phase_idx} ({'busy' if busy_phase else 'idle'}): "
               f"{ticks} ticks, {scaled} scale actions, rss={rss} KiB fds={fds}", flush=True)
@@ -93,11 +95,11 @@ finally:
         pass
     exporter.terminate(); exporter.wait()
     core.otlp_shutdown()
-    spans = len(collector.span_names())
+    spans = spans_total[0] + collector.drain()[0]
     collector.stop()
     core.informers_reset()
     sampler.stop(); backend.stop()
-    print(f"otlp spans exported during soak: {spans}", flush=True)
+    print(f"otlp trace export batches during soak: {spans}", flush=True)
 
 for phase, ticks, scaled in phases:
     if phase == "busy" and scaled != 0:
