@@ -75,6 +75,8 @@ class FakeApiServer:
         self._event_cv = threading.Condition(self._lock)
         self._closing = False
         self.watch_requests = 0  # watch connections served (test observability)
+        # failure injection for informer resilience tests
+        self.watch_410_next = 0  # respond 410 Gone to the next N watch requests
 
         fixture = self
 
@@ -148,6 +150,14 @@ class FakeApiServer:
                 deadline = time.monotonic() + timeout_s
                 with fixture._lock:
                     fixture.watch_requests += 1
+                    if fixture.watch_410_next > 0:
+                        fixture.watch_410_next -= 1
+                        gone = True
+                    else:
+                        gone = False
+                if gone:  # resourceVersion too old: client must re-LIST
+                    return self._send(410, self._status(
+                        410, "Expired", "too old resource version"))
                 self.send_response(200)
                 self.send_header("Content-Type", "application/json")
                 self.send_header("Transfer-Encoding", "chunked")
@@ -454,3 +464,16 @@ class FakeApiServer:
             "status": {"phase": phase},
         }
         return self.put("Pod", pod)
+
+    def inject_watch_error(self, kind: str, ns: str):
+        """Append an in-band ERROR watch event (the apiserver's 410-inside-
+        the-stream form); watchers must invalidate and re-LIST."""
+        with self._lock:
+            self._rv += 1
+            self._events.append((self._rv, "ERROR", kind, ns,
+                                 {"kind": "Status", "apiVersion": "v1",
+                                  "status": "Failure", "reason": "Expired",
+                                  "code": 410,
+                                  "metadata": {"resourceVersion": str(self._rv),
+                                               "name": "", "namespace": ns}}))
+            self._event_cv.notify_all()

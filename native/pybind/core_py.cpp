@@ -260,6 +260,27 @@ PYBIND11_MODULE(_pruner_core, m) {
 
   m.def("get_prometheus_token", [] { return get_prometheus_token(); });
 
+  m.def("http_stream_lines",
+        [](const std::string& url) {
+          auto parsed = http::Url::parse(url);
+          if (!parsed) throw std::runtime_error("bad url");
+          std::vector<std::string> lines;
+          {
+            py::gil_scoped_release nogil;
+            http::Client client(*parsed, http::ClientOptions{});
+            http::Request r;
+            r.path = parsed->path;
+            auto stream = client.open_stream(r);
+            std::string line;
+            while (stream->read_line(&line)) lines.push_back(line);
+          }
+          py::list out;
+          for (auto& l : lines) out.append(l);
+          return out;
+        },
+        "Open a streaming GET and return every decoded body line "
+        "(BodyStream chunk-decoder test surface)");
+
   m.def("informers_reset",
         [] {
           py::gil_scoped_release nogil;

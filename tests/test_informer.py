@@ -154,3 +154,70 @@ def test_watch_strategy_daemon_binary_e2e(pruner_bin, fake_api, fake_prom):
         capture_output=True, text=True, timeout=60, env=env)
     assert r.returncode == 0, r.stderr
     assert fake_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0
+
+
+def test_watch_410_gone_triggers_relist_and_recovers(core, monkeypatch,
+                                                     informer_reset):
+    """resourceVersion too old (HTTP 410 on the watch request): the informer
+    re-LISTs and decisions stay correct."""
+    monkeypatch.setenv("PROMETHEUS_TOKEN", "t")
+    with FakeApiServer() as api, FakePrometheus() as prom:
+        dep = api.add_deployment("d", "ml")
+        rs = api.add_replicaset("d-rs", "ml", owner=dep)
+        api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                    owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+        prom.add_idle_series("p0", "ml")
+        monkeypatch.setenv("GPU_PRUNER_K8S_URL", api.url)
+        cfg = _cfg(prom.url, "watch")
+        out1 = core.run_tick(cfg)
+        assert out1["shutdown_events"] == 1
+        lists_before = len(_list_requests(api))
+
+        # break the live pods stream (ERROR) so it reopens NOW, and answer
+        # the reopen with 410 Gone — exercising both invalidation paths
+        api.watch_410_next = 2
+        api.inject_watch_error("Pod", "ml")
+        # new state the informer can only learn after recovering
+        api.add_pod("p1", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                    owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+        prom.add_idle_series("p1", "ml")
+        deadline = time.monotonic() + 10
+        out = None
+        while time.monotonic() < deadline:
+            out = core.run_tick(cfg)
+            if out["num_unique_pods"] == 2:
+                break
+            time.sleep(0.2)
+        assert out["num_unique_pods"] == 2, out
+        assert out["shutdown_events"] == 1
+        # at least one collection actually re-LISTed after its 410
+        assert len(_list_requests(api)) > lists_before
+
+
+def test_watch_inband_error_event_triggers_relist(core, monkeypatch,
+                                                  informer_reset):
+    """An ERROR event inside the stream (410 delivered in-band) invalidates
+    the store; the informer re-LISTs and keeps deciding correctly."""
+    monkeypatch.setenv("PROMETHEUS_TOKEN", "t")
+    with FakeApiServer() as api, FakePrometheus() as prom:
+        dep = api.add_deployment("d", "ml")
+        rs = api.add_replicaset("d-rs", "ml", owner=dep)
+        api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                    owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+        prom.add_idle_series("p0", "ml")
+        monkeypatch.setenv("GPU_PRUNER_K8S_URL", api.url)
+        cfg = _cfg(prom.url, "watch")
+        assert core.run_tick(cfg)["shutdown_events"] == 1
+        lists_before = len(_list_requests(api))
+
+        api.inject_watch_error("Pod", "ml")
+        deadline = time.monotonic() + 10
+        relisted = False
+        while time.monotonic() < deadline:
+            out = core.run_tick(cfg)
+            assert out["shutdown_events"] == 1  # decisions never degrade
+            if len(_list_requests(api)) > lists_before:
+                relisted = True
+                break
+            time.sleep(0.2)
+        assert relisted, "ERROR event did not force a re-LIST"
