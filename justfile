@@ -26,6 +26,11 @@ test-all: test
 test-gpu:
     python3 -m pytest tests -q -m gpu
 
+# real-apiserver tier: point KUBEBUILDER_ASSETS at a dir containing etcd +
+# kube-apiserver (setup-envtest use -p path); skips cleanly otherwise
+test-envtest:
+    python3 -m pytest tests/test_envtest_e2e.py -q -rs
+
 bench:
     python3 bench.py --steps 20 --warmup 3
 
