@@ -249,6 +249,16 @@ def test_owner_walk_scales_deployment_not_replicaset(real_apiserver, fake_prom):
     names = [e["metadata"]["name"] for e in events.get("items", [])]
     assert any(n.startswith("gpuscaler-") for n in names), names
 
+    # the watch strategy against a REAL apiserver: informers LIST + watch
+    # genuine streams; a repeat tick is idempotent (replicas stay 0)
+    r2 = _run_pruner(real_apiserver, fake_prom, "-t", "1", "--grace-period", "0",
+                     "--eval-strategy", "watch")
+    assert r2.returncode == 0, r2.stderr
+    status, dep_after2, _ = http_request(
+        base, "GET", "/apis/apps/v1/namespaces/ml/deployments/e2e-dep",
+        headers=headers, insecure=True)
+    assert dep_after2["spec"]["replicas"] == 0
+
 
 def test_notebook_and_inferenceservice_crd_paths(real_apiserver, fake_prom):
     """Reference gap closed (kind e2e never installed the CRDs): Notebook
