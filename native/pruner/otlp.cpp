@@ -63,6 +63,38 @@ struct ThreadCtx {
 };
 thread_local ThreadCtx t_ctx;
 
+// Finished spans land in PER-THREAD buffers (a single global mutex was the
+// hot path: ~3k spans per 1000-pod tick across the worker pool). Buffers
+// register once in a global list the exporter drains; shared_ptr keeps a
+// buffer alive past thread exit so its tail still exports.
+struct ThreadSpanBuf {
+  std::mutex mu;  // uncontended except while the exporter drains
+  std::vector<FinishedSpan> spans;
+};
+
+struct BufRegistry {
+  std::mutex mu;
+  std::vector<std::shared_ptr<ThreadSpanBuf>> bufs;
+};
+
+BufRegistry& buf_registry() {
+  static BufRegistry r;
+  return r;
+}
+
+ThreadSpanBuf& thread_buf() {
+  thread_local std::shared_ptr<ThreadSpanBuf> buf = [] {
+    auto b = std::make_shared<ThreadSpanBuf>();
+    auto& reg = buf_registry();
+    std::lock_guard<std::mutex> lock(reg.mu);
+    reg.bufs.push_back(b);
+    return b;
+  }();
+  return *buf;
+}
+
+constexpr size_t kMaxPerThreadSpans = 16384;  // drop-oldest beyond this
+
 State& state() {
   static State s;
   return s;
@@ -246,6 +278,22 @@ void export_once() {
   {
     std::lock_guard<std::mutex> lock(s.mu);
     all.swap(s.spans);
+  }
+  {
+    // drain every thread's buffer
+    std::vector<std::shared_ptr<ThreadSpanBuf>> bufs;
+    {
+      auto& reg = buf_registry();
+      std::lock_guard<std::mutex> lock(reg.mu);
+      bufs = reg.bufs;
+    }
+    for (auto& b : bufs) {
+      std::lock_guard<std::mutex> lock(b->mu);
+      if (b->spans.empty()) continue;
+      all.insert(all.end(), std::make_move_iterator(b->spans.begin()),
+                 std::make_move_iterator(b->spans.end()));
+      b->spans.clear();
+    }
   }
   for (size_t base = 0; base < all.size(); base += kMaxSpansPerPost) {
     size_t n = std::min(kMaxSpansPerPost, all.size() - base);
@@ -441,16 +489,16 @@ SpanGuard::~SpanGuard() {
   if (span_id_.empty()) return;  // was created before/without otlp::init
   if (!t_ctx.stack.empty() && t_ctx.stack.back() == span_id_) t_ctx.stack.pop_back();
   if (!enabled()) return;
-  State& s = state();
-  std::lock_guard<std::mutex> lock(s.mu);
-  if (s.spans.size() >= kMaxBufferedSpans) {
+  ThreadSpanBuf& buf = thread_buf();
+  std::lock_guard<std::mutex> lock(buf.mu);
+  if (buf.spans.size() >= kMaxPerThreadSpans) {
     // bounded buffer: a stalled exporter drops oldest spans instead of
     // growing the daemon heap without limit
-    s.spans.erase(s.spans.begin(),
-                  s.spans.begin() + static_cast<long>(kMaxBufferedSpans / 10));
-    s.dropped_spans.fetch_add(kMaxBufferedSpans / 10, std::memory_order_relaxed);
+    buf.spans.erase(buf.spans.begin(),
+                    buf.spans.begin() + static_cast<long>(kMaxPerThreadSpans / 10));
+    state().dropped_spans.fetch_add(kMaxPerThreadSpans / 10, std::memory_order_relaxed);
   }
-  s.spans.push_back({name_, trace_id_, span_id_, parent_id_, start_ns_, now_unix_ns()});
+  buf.spans.push_back({name_, trace_id_, span_id_, parent_id_, start_ns_, now_unix_ns()});
 }
 
 ContextGuard::ContextGuard(const SpanContext& parent) {
