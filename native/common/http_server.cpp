@@ -16,6 +16,7 @@
 #include <stdexcept>
 
 #include "strutil.hpp"
+#include "tsan_compat.hpp"
 
 namespace http {
 
@@ -45,7 +46,7 @@ struct ConnRegistry {
   }
   void wait_drained(int timeout_ms) {
     std::unique_lock<std::mutex> lock(mu);
-    cv.wait_for(lock, std::chrono::milliseconds(timeout_ms), [&] { return active == 0; });
+    qx::cv_wait_for(cv, lock, std::chrono::milliseconds(timeout_ms), [&] { return active == 0; });
   }
 };
 

@@ -4,6 +4,7 @@
 
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
+#include "../common/tsan_compat.hpp"
 
 namespace pruner {
 
@@ -44,8 +45,8 @@ void Informer::stop() {
 
 bool Informer::wait_synced(int timeout_ms) {
   std::unique_lock<std::mutex> lock(mu_);
-  cv_.wait_for(lock, std::chrono::milliseconds(timeout_ms),
-               [this] { return synced_ || stop_.load(); });
+  qx::cv_wait_for(cv_, lock, std::chrono::milliseconds(timeout_ms),
+                  [this] { return synced_ || stop_.load(); });
   return synced_;
 }
 
@@ -178,7 +179,7 @@ void Informer::run() {
       if (!watchable) {
         // nothing to watch (collection absent): poll the LIST occasionally
         std::unique_lock<std::mutex> lock(mu_);
-        cv_.wait_for(lock, std::chrono::seconds(5), [this] { return stop_.load(); });
+        qx::cv_wait_for(cv_, lock, std::chrono::seconds(5), [this] { return stop_.load(); });
         if (stop_) return;
         synced_ = false;  // retry the LIST (the CRD may have appeared)
       }
@@ -192,7 +193,7 @@ void Informer::run() {
       LOGW(TARGET, "watch loop for " + path_ + " failed (" + e.what() + "), retrying in " +
                        std::to_string(backoff_ms) + " ms");
       std::unique_lock<std::mutex> lock(mu_);
-      cv_.wait_for(lock, std::chrono::milliseconds(backoff_ms), [this] { return stop_.load(); });
+      qx::cv_wait_for(cv_, lock, std::chrono::milliseconds(backoff_ms), [this] { return stop_.load(); });
       if (stop_) return;
       backoff_ms = std::min(backoff_ms * 2, 10000);
     }

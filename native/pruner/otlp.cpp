@@ -15,6 +15,7 @@
 #include "../common/json.hpp"
 #include "../common/log.hpp"
 #include "../common/strutil.hpp"
+#include "../common/tsan_compat.hpp"
 
 namespace otlp {
 
@@ -438,7 +439,7 @@ void init(const std::string& service_name) {
   s.exporter = std::thread([&s] {
     std::unique_lock<std::mutex> lock(s.mu);
     while (!s.stop) {
-      s.cv.wait_for(lock, std::chrono::milliseconds(s.interval_ms));
+      qx::cv_wait_for(s.cv, lock, std::chrono::milliseconds(s.interval_ms));
       if (s.stop) break;
       lock.unlock();
       export_once();

@@ -7,6 +7,7 @@
 #include <cstdio>
 
 #include "../common/strutil.hpp"
+#include "../common/tsan_compat.hpp"
 #include "resources.hpp"
 
 namespace pruner {
@@ -300,7 +301,7 @@ http::ServerResponse SyntheticBackend::handle_k8s(const http::ServerRequest& req
           bookmark_sent = true;
           return true;  // deliver the bookmark; next call ends the stream
         }
-        event_cv_.wait_for(lock, std::chrono::milliseconds(100));
+        qx::cv_wait_for(event_cv_, lock, std::chrono::milliseconds(100));
       }
     };
     return resp;
