@@ -17,10 +17,17 @@ def test_daemon_mode_ticks_on_interval(pruner_bin, fake_api, fake_prom):
          "--check-interval", "1"],
         env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
     try:
-        time.sleep(3.5)
+        # wait until at least 2 ticks landed (loaded CI boxes can stretch
+        # startup); then verify the cadence upper bound over the window
+        deadline = time.monotonic() + 15
+        t0 = time.monotonic()
+        while time.monotonic() < deadline and len(fake_prom.queries) < 2:
+            time.sleep(0.1)
+        elapsed = time.monotonic() - t0
         assert p.poll() is None, "daemon should keep running"
         n = len(fake_prom.queries)
-        assert 2 <= n <= 6, f"expected ~3 ticks at 1s interval, saw {n}"
+        assert n >= 2, f"expected >=2 ticks at 1s interval within 15s, saw {n}"
+        assert n <= elapsed + 3, f"{n} ticks in {elapsed:.1f}s — interval ignored?"
     finally:
         p.kill()
         p.wait()
@@ -167,10 +174,14 @@ def test_latency_injected_tick_uses_concurrency(core, monkeypatch):
                            "prometheus_url": b.prom_url, "max_concurrency": 32,
                            "eval_strategy": "get"})
         core.run_tick(cfg)  # warmup
-        t0 = _time.perf_counter()
-        out = core.run_tick(cfg)
-        dt = _time.perf_counter() - t0
+        # best of 3 to shrug off CI load spikes; a serial engine needs
+        # >= 1.2 s regardless
+        dt = 10.0
+        for _ in range(3):
+            t0 = _time.perf_counter()
+            out = core.run_tick(cfg)
+            dt = min(dt, _time.perf_counter() - t0)
         assert out["num_unique_pods"] == 200
-        assert dt < 0.8, f"tick took {dt:.2f}s — concurrency regression?"
+        assert dt < 1.0, f"tick took {dt:.2f}s — concurrency regression?"
     finally:
         b.stop()
