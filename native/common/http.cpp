@@ -14,6 +14,7 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cstring>
 
 #include "strutil.hpp"
@@ -349,7 +350,8 @@ public:
   ClientOptions opts_;
   int fd_ = -1;
   SSL* ssl_ = nullptr;
-  bool broken_ = false;
+  // written by shutdown_socket() from other threads (watch-stream abort)
+  std::atomic<bool> broken_{false};
 };
 
 namespace {
