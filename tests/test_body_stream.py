@@ -103,3 +103,38 @@ def test_content_length_stream(core):
             b"Content-Length: " + str(len(body)).encode() + b"\r\n\r\n")
     addr = serve_fragments([head + body[:5], body[5:]])
     assert stream_lines(core, addr) == ['{"a":1}', '{"b":2}']
+
+
+# ---- property-based fuzz: random payloads x random chunking x random
+# fragmentation must always round-trip ------------------------------------
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import HealthCheck, given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=60, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(
+    lines=st.lists(st.text(alphabet=st.characters(min_codepoint=0x20,
+                                                  max_codepoint=0x7E),
+                           min_size=1, max_size=80),
+                   min_size=1, max_size=12),
+    chunk_sizes=st.lists(st.integers(min_value=1, max_value=64), min_size=1,
+                         max_size=8),
+    frag=st.integers(min_value=1, max_value=97),
+)
+def test_chunk_decoder_roundtrip_fuzz(core, lines, chunk_sizes, frag):
+    body = "".join(l + "\n" for l in lines).encode()
+    wire = b""
+    i = 0
+    k = 0
+    while i < len(body):
+        n = min(chunk_sizes[k % len(chunk_sizes)], len(body) - i)
+        wire += chunked(body[i:i + n])
+        i += n
+        k += 1
+    wire += b"0\r\n\r\n"
+    whole = HEAD + wire
+    frags = [whole[i:i + frag] for i in range(0, len(whole), frag)]
+    addr = serve_fragments(frags)
+    assert stream_lines(core, addr) == lines
