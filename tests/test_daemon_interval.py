@@ -96,7 +96,14 @@ def test_sigterm_graceful_shutdown(pruner_bin, fake_api, fake_prom):
 
 def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
     """--metrics-port serves the six counters + /healthz (MI355X-native add)."""
+    import socket
     import urllib.request
+
+    # ephemeral port: fixed ports collide when suites run in parallel
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
 
     dep = fake_api.add_deployment("d", "ml")
     rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
@@ -109,7 +116,7 @@ def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
     p = subprocess.Popen(
         [pruner_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
          "--check-interval", "1", "--run-mode", "scale-down",
-         "--metrics-port", "19490"],
+         "--metrics-port", str(port)],
         env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
     try:
         text = None
@@ -117,7 +124,7 @@ def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
             time.sleep(0.2)
             try:
                 text = urllib.request.urlopen(
-                    "http://127.0.0.1:19490/metrics", timeout=2).read().decode()
+                    f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
                 if "gpu_pruner_query_successes_total" in text:
                     break
             except OSError:
@@ -125,7 +132,7 @@ def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
         assert text and "gpu_pruner_query_successes_total" in text
         assert "gpu_pruner_scale_successes_total" in text
         health = urllib.request.urlopen(
-            "http://127.0.0.1:19490/healthz", timeout=2).read()
+            f"http://127.0.0.1:{port}/healthz", timeout=2).read()
         assert health == b"ok\n"
     finally:
         p.terminate()
