@@ -20,6 +20,10 @@ from gpu_pruner_amd.fixtures import FakeApiServer, FakePrometheus, build_synthet
 
 @pytest.fixture
 def informer_reset(core):
+    # reset BEFORE too: ephemeral ports get reused across fixtures, and a
+    # stale informer keyed on a recycled (url, path) from a previous test
+    # could alias the new fixture
+    core.informers_reset()
     yield
     core.informers_reset()
 
