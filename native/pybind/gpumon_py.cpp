@@ -8,6 +8,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "../common/grpc_client.hpp"
 #include "../common/json.hpp"
 #include "../common/log.hpp"
 #include "../exporter/attrib.hpp"
@@ -79,6 +80,7 @@ PYBIND11_MODULE(_gpumon, m) {
   logx::init(logx::Format::Default);
 
   py::register_exception<SamplerError>(m, "SamplerError");
+  py::register_exception<grpcx::GrpcError>(m, "GrpcError");
 
   py::class_<Sampler>(m, "Sampler")
       .def(py::init<int, double, double>(), py::arg("poll_interval_ms") = 1000,
@@ -144,6 +146,25 @@ PYBIND11_MODULE(_gpumon, m) {
         "(hand-rolled h2c + protobuf, no grpc library)");
 
   py::register_exception<PodResourcesError>(m, "PodResourcesError");
+
+  m.def("grpc_unary_call",
+        [](const std::string& host, uint16_t port, const std::string& method_path,
+           py::bytes request_msg, int timeout_ms) -> py::bytes {
+          std::string req = request_msg;
+          std::string resp;
+          {
+            py::gil_scoped_release nogil;
+            grpcx::Target t;
+            t.host = host;
+            t.port = port;
+            t.authority = host + ":" + std::to_string(port);
+            resp = grpcx::unary_call(t, method_path, req, timeout_ms);
+          }
+          return py::bytes(resp);
+        },
+        py::arg("host"), py::arg("port"), py::arg("method_path"),
+        py::arg("request_msg"), py::arg("timeout_ms") = 5000,
+        "Raw unary gRPC over h2c (native/common/grpc_client.cpp test surface)");
 
   m.def("decode_list_response", [](py::bytes payload) {
     std::string data = payload;
