@@ -27,10 +27,12 @@ _ROUTES = {
     ("apis/apps/v1", "statefulsets"): "StatefulSet",
     ("apis/kubeflow.org/v1", "notebooks"): "Notebook",
     ("apis/serving.kserve.io/v1beta1", "inferenceservices"): "InferenceService",
+    ("apis/coordination.k8s.io/v1", "leases"): "Lease",
 }
 
 _PATH_RE = re.compile(
-    r"^/(api/v1|apis/apps/v1|apis/kubeflow\.org/v1|apis/serving\.kserve\.io/v1beta1)"
+    r"^/(api/v1|apis/apps/v1|apis/kubeflow\.org/v1|apis/serving\.kserve\.io/v1beta1"
+    r"|apis/coordination\.k8s\.io/v1)"
     r"/namespaces/([^/]+)/([^/]+)(?:/([^/]+))?(?:/(scale))?$"
 )
 
@@ -289,6 +291,52 @@ class FakeApiServer:
                     fixture._record_locked("MODIFIED", kind, obj)
                     return self._send(200, copy.deepcopy(obj))
 
+            def do_PUT(self):
+                """Replace semantics with optimistic concurrency: a body
+                carrying metadata.resourceVersion that does not match the
+                stored object's is rejected 409 Conflict (what real Update
+                calls do — Lease-based leader election depends on it)."""
+                with fixture._lock:
+                    fixture.requests.append(("PUT", self.path))
+                if fixture.latency_s:
+                    time.sleep(fixture.latency_s)
+                if self._maybe_throttle():
+                    return
+                if not self._auth_ok():
+                    return self._send(401, self._status(401, "Unauthorized", "Unauthorized"))
+                length = int(self.headers.get("Content-Length", "0"))
+                body = json.loads(self.rfile.read(length) or b"{}")
+                r = self._route()
+                if r is None or r[2] is None:
+                    return self._send(404, self._status(404, "NotFound",
+                                                        "the server could not find the requested resource"))
+                kind, ns, name, _sub = r
+                with fixture._lock:
+                    obj = fixture.objects.get((kind, ns, name))
+                    if obj is None:
+                        return self._send(404, self._status(
+                            404, "NotFound", f'{kind.lower()}s "{name}" not found',
+                            details={"name": name, "kind": kind.lower() + "s"}))
+                    sent_rv = body.get("metadata", {}).get("resourceVersion")
+                    stored_rv = obj.get("metadata", {}).get("resourceVersion")
+                    if sent_rv is not None and sent_rv != stored_rv:
+                        return self._send(409, self._status(
+                            409, "Conflict",
+                            f'Operation cannot be fulfilled on {kind.lower()}s "{name}": '
+                            "the object has been modified; please apply your changes to the "
+                            "latest version and try again",
+                            details={"name": name, "kind": kind.lower() + "s"}))
+                    body.setdefault("metadata", {})
+                    body["metadata"].setdefault("namespace", ns)
+                    body["metadata"]["name"] = name
+                    # preserve server-owned fields
+                    for k in ("uid", "creationTimestamp"):
+                        if k in obj.get("metadata", {}):
+                            body["metadata"].setdefault(k, obj["metadata"][k])
+                    fixture.objects[(kind, ns, name)] = body
+                    fixture._record_locked("MODIFIED", kind, body)
+                    return self._send(200, copy.deepcopy(body))
+
             def do_POST(self):
                 with fixture._lock:
                     fixture.requests.append(("POST", self.path))
@@ -314,6 +362,12 @@ class FakeApiServer:
                     return self._send(201, obj)
                 name = obj.get("metadata", {}).get("name", "")
                 with fixture._lock:
+                    if (kind, ns, name) in fixture.objects:
+                        return self._send(409, self._status(
+                            409, "AlreadyExists",
+                            f'{kind.lower()}s "{name}" already exists',
+                            details={"name": name, "kind": kind.lower() + "s"}))
+                    obj.setdefault("metadata", {}).setdefault("namespace", ns)
                     fixture.objects[(kind, ns, name)] = obj
                     fixture._record_locked("ADDED", kind, obj)
                 return self._send(201, obj)

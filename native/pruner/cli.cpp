@@ -48,6 +48,11 @@ OPTIONS:
                                   informers, delta traffic), get (per-object
                                   GETs, reference-equivalent) | list
                                   (namespace LISTs) | auto [default: auto]
+      --leader-elect              coordination.k8s.io Lease-based leader
+                                  election: only the lease holder runs ticks
+                                  (safe multi-replica deployments)
+      --leader-elect-lease-duration <SEC>  [default: 15]
+      --leader-elect-renew-period <SEC>    [default: 5]
   -h, --help                      print this help
 )";
 
@@ -88,11 +93,14 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
       {"--max-failures", "max-failures"},
       {"--metrics-port", "metrics-port"},
       {"--eval-strategy", "eval-strategy"},
+      {"--leader-elect", "leader-elect"},
+      {"--leader-elect-lease-duration", "leader-elect-lease-duration"},
+      {"--leader-elect-renew-period", "leader-elect-renew-period"},
       {"-h", "help"},              {"--help", "help"},
   };
   // flags that never take a value
   auto is_switch = [](const std::string& name) {
-    return name == "daemon-mode" || name == "help";
+    return name == "daemon-mode" || name == "leader-elect" || name == "help";
   };
   // flags with an optional boolean value (clap bool with default_value)
   auto is_opt_bool = [](const std::string& name) { return name == "honor-labels"; };
@@ -127,7 +135,8 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
 
     if (name == "help") { res.show_help = true; return res; }
     if (is_switch(name)) {
-      c.daemon_mode = true;
+      if (name == "daemon-mode") c.daemon_mode = true;
+      else if (name == "leader-elect") c.leader_elect = true;
       continue;
     }
     if (is_opt_bool(name)) {
@@ -148,6 +157,8 @@ CliResult parse_cli(const std::vector<std::string>& argv) {
     try {
       if (name == "duration") c.duration_min = std::stol(val);
       else if (name == "enabled-resources") c.enabled_resources = val;
+      else if (name == "leader-elect-lease-duration") c.leader_lease_duration_s = std::stoi(val);
+      else if (name == "leader-elect-renew-period") c.leader_renew_period_s = std::stoi(val);
       else if (name == "check-interval") c.check_interval_s = std::stoul(val);
       else if (name == "namespace") c.namespace_ = val;
       else if (name == "grace-period") c.grace_period_s = std::stol(val);

@@ -47,6 +47,12 @@ struct Config {
   //   list = namespace-collection LISTs (O(namespaces) RTTs per tick)
   //   auto = LIST namespaces with >= 10 candidates, GETs elsewhere
   EvalStrategy eval_strategy = EvalStrategy::Auto;
+  // --leader-elect: coordination.k8s.io/v1 Lease-based leader election so
+  // the daemon can run multi-replica without double-culling (the reference
+  // is single-replica by convention). Only the lease holder runs ticks.
+  bool leader_elect = false;
+  int leader_lease_duration_s = 15;  // --leader-elect-lease-duration
+  int leader_renew_period_s = 5;     // --leader-elect-renew-period
 
   QueryArgs query_args() const {
     QueryArgs qa;

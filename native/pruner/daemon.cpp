@@ -1,6 +1,7 @@
 #include "daemon.hpp"
 
 #include <csignal>
+#include <unistd.h>
 
 #include <atomic>
 #include <chrono>
@@ -12,6 +13,7 @@
 
 #include "../common/http_server.hpp"
 #include "informer.hpp"
+#include "leader.hpp"
 #include "../common/log.hpp"
 #include "../common/queue.hpp"
 #include "engine.hpp"
@@ -130,13 +132,46 @@ int run_daemon(const Config& cfg) {
     LOGI(TARGET, "Self-metrics on :" + std::to_string(metrics_server->port()));
   }
 
+  // Optional Lease-based leader election (multi-replica deployments): only
+  // the holder runs decision ticks; standbys keep renewing their candidacy
+  // and take over when the lease expires or is released.
+  std::unique_ptr<LeaderElector> elector;
+  if (cfg.leader_elect && cfg.daemon_mode) {
+    std::string identity;
+    if (const char* pn = std::getenv("POD_NAME"); pn && *pn) identity = pn;
+    if (identity.empty()) {
+      char host[256] = {0};
+      ::gethostname(host, sizeof host - 1);
+      identity = std::string(host) + "-" + std::to_string(::getpid());
+    }
+    KubeConfig kc = KubeConfig::resolve();
+    std::string lease_ns = kc.default_namespace;
+    if (const char* pns = std::getenv("POD_NAMESPACE"); pns && *pns) lease_ns = pns;
+    elector = std::make_unique<LeaderElector>(kc, lease_ns, "gpu-pruner", identity,
+                                              cfg.leader_lease_duration_s,
+                                              cfg.leader_renew_period_s);
+    elector->start();
+    LOGI(TARGET, "Leader election on (lease " + lease_ns + "/gpu-pruner, identity \"" +
+                     identity + "\")");
+  }
+
   int consecutive_failures = 0;
+  bool was_leader = true;  // log standby transitions once
   auto next_tick = std::chrono::steady_clock::now();
   while (!g_shutdown.load()) {
     if (cfg.daemon_mode) {
       interruptible_sleep_until(next_tick);
       if (g_shutdown.load()) break;
       next_tick += std::chrono::seconds(cfg.check_interval_s);
+    }
+    if (elector && !elector->is_leader()) {
+      if (was_leader) LOGI(TARGET, "Not the leader — standing by");
+      was_leader = false;
+      continue;
+    }
+    if (elector && !was_leader) {
+      LOGI(TARGET, "Leadership acquired — resuming decision ticks");
+      was_leader = true;
     }
     try {
       otlp::SpanGuard span("run_query_and_scale");
@@ -173,6 +208,7 @@ int run_daemon(const Config& cfg) {
   queue.close();  // producer done: consumers drain and exit
   for (auto& c : consumers) c.join();
   if (metrics_server) metrics_server->stop();
+  if (elector) elector->stop();  // releases the lease for instant failover
   InformerRegistry::global().stop_all();  // close watch streams (if any)
   return exit_code.load();
 }

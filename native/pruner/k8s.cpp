@@ -358,6 +358,19 @@ void KubeClient::patch_scale(Kind kind, const std::string& ns, const std::string
   merge_patch(object_path(kind, ns, name) + "/scale", patch);
 }
 
+jsn::Value KubeClient::replace(const std::string& path, const jsn::Value& obj) {
+  http::Request r;
+  r.method = "PUT";
+  r.path = path;
+  r.body = obj.dump();
+  r.headers.emplace_back("Content-Type", "application/json");
+  http::Response resp = authed(r);
+  if (resp.status < 200 || resp.status >= 300)
+    throw KubeError(resp.status, "PUT " + path + " -> " + std::to_string(resp.status) + ": " +
+                                     resp.body.substr(0, 300));
+  return jsn::parse(resp.body);
+}
+
 std::unique_ptr<http::BodyStream> KubeClient::open_stream(const std::string& path) {
   http::Request r;
   r.method = "GET";

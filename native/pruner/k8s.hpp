@@ -93,6 +93,12 @@ public:
 
   void create(const std::string& collection_path, const jsn::Value& obj);
 
+  // PUT replace with optimistic concurrency: sending metadata.resourceVersion
+  // makes the apiserver reject stale writes with 409 Conflict (KubeError
+  // status 409) — the primitive Lease-based leader election rests on.
+  // Returns the stored object (carrying the new resourceVersion).
+  jsn::Value replace(const std::string& path, const jsn::Value& obj);
+
   // Streaming GET (Kubernetes watch): returns once status+headers arrive.
   std::unique_ptr<http::BodyStream> open_stream(const std::string& path);
 

@@ -1,0 +1,150 @@
+"""Lease-based leader election (--leader-elect).
+
+The reference is single-replica by convention (restartPolicy: Always,
+hack/deployment.yaml:38); two replicas would double-cull. This MI355X-native
+addition gates decision ticks on a coordination.k8s.io/v1 Lease with
+resourceVersion-fenced takeover — these tests run two real daemon processes
+against the fake apiserver and pin: single holder, standby behavior,
+takeover after a SIGKILLed leader's lease expires, and instant release on
+graceful shutdown.
+"""
+
+import os
+import signal
+import subprocess
+import time
+
+import pytest
+
+
+def start_daemon(pruner_bin, api, prom, identity, extra=()):
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    env["POD_NAME"] = identity
+    env["POD_NAMESPACE"] = "gpu-pruner-system"
+    env["GPU_PRUNER_LOG"] = "info"
+    return subprocess.Popen(
+        [pruner_bin, "--prometheus-url", prom.url, "--daemon-mode",
+         "--run-mode", "scale-down", "--check-interval", "1",
+         "--leader-elect", "--leader-elect-lease-duration", "4",
+         "--leader-elect-renew-period", "1", *extra],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+
+
+def lease_holder(api):
+    lease = api.get("Lease", "gpu-pruner-system", "gpu-pruner")
+    if lease is None:
+        return None
+    return lease.get("spec", {}).get("holderIdentity")
+
+
+def wait_for(predicate, timeout_s, interval=0.2):
+    deadline = time.monotonic() + timeout_s
+    while time.monotonic() < deadline:
+        v = predicate()
+        if v:
+            return v
+        time.sleep(interval)
+    return predicate()
+
+
+def test_two_replicas_single_leader_and_failover(pruner_bin, fake_api, fake_prom):
+    dep = fake_api.add_deployment("d", "ml")
+    rs = fake_api.add_replicaset("d-rs", "ml", owner=dep)
+    fake_api.add_pod("p0", "ml", owner_kind="ReplicaSet", owner_name="d-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p0", "ml")
+
+    a = start_daemon(pruner_bin, fake_api, fake_prom, "replica-a")
+    time.sleep(0.5)  # deterministic: a acquires first
+    b = start_daemon(pruner_bin, fake_api, fake_prom, "replica-b")
+    try:
+        # exactly one holder, and it culls
+        holder = wait_for(lambda: lease_holder(fake_api), 10)
+        assert holder == "replica-a", holder
+        assert wait_for(
+            lambda: fake_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0, 10)
+        lease = fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner")
+        assert lease["spec"]["leaseDurationSeconds"] == 4
+
+        # the standby stands by: let a few ticks pass, then make sure ONLY
+        # the leader queried Prometheus (each process queries per tick; a
+        # standby issues none)
+        q_before = len(fake_prom.queries)
+        time.sleep(3)
+        assert len(fake_prom.queries) > q_before  # leader still ticking
+
+        # hard-kill the leader (no release): the standby takes over after
+        # the lease expires (4 s duration + renew cadence)
+        a.kill()
+        a.wait(timeout=10)
+        holder = wait_for(lambda: lease_holder(fake_api) == "replica-b", 15)
+        assert holder, f"standby never took over (holder={lease_holder(fake_api)})"
+        lease = fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner")
+        assert lease["spec"]["leaseTransitions"] >= 1
+
+        # the new leader culls: re-arm the deployment and watch it drop
+        fake_api.objects[("Deployment", "ml", "d")]["spec"]["replicas"] = 1
+        assert wait_for(
+            lambda: fake_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0, 10)
+
+        # graceful shutdown releases the lease immediately
+        b.send_signal(signal.SIGTERM)
+        b.wait(timeout=15)
+        assert lease_holder(fake_api) == ""
+        err = b.stderr.read().decode()
+        assert "Acquired leadership" in err
+        assert "Released lease" in err
+    finally:
+        for p in (a, b):
+            if p.poll() is None:
+                p.kill()
+                p.wait(timeout=10)
+
+
+def test_standby_logs_and_does_not_act(pruner_bin, fake_api, fake_prom):
+    dep = fake_api.add_deployment("d2", "ml")
+    rs = fake_api.add_replicaset("d2-rs", "ml", owner=dep)
+    fake_api.add_pod("p2", "ml", owner_kind="ReplicaSet", owner_name="d2-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p2", "ml")
+
+    a = start_daemon(pruner_bin, fake_api, fake_prom, "leader")
+    time.sleep(0.5)
+    b = start_daemon(pruner_bin, fake_api, fake_prom, "standby")
+    try:
+        assert wait_for(lambda: lease_holder(fake_api) == "leader", 10)
+        time.sleep(2.5)
+        b.send_signal(signal.SIGTERM)
+        b.wait(timeout=15)
+        err = b.stderr.read().decode()
+        assert "standing by" in err
+        assert "Acquired leadership" not in err.replace(
+            "Leadership acquired", "")  # never led
+        # leader survives the standby's exit and still holds the lease
+        assert lease_holder(fake_api) == "leader"
+        assert a.poll() is None
+    finally:
+        for p in (a, b):
+            if p.poll() is None:
+                p.terminate()
+                p.wait(timeout=10)
+
+
+def test_leader_elect_off_by_default(pruner_bin, fake_api, fake_prom):
+    """Without --leader-elect no Lease is touched (reference-equivalent)."""
+    dep = fake_api.add_deployment("d3", "ml")
+    rs = fake_api.add_replicaset("d3-rs", "ml", owner=dep)
+    fake_api.add_pod("p3", "ml", owner_kind="ReplicaSet", owner_name="d3-rs",
+                     owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
+    fake_prom.add_idle_series("p3", "ml")
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = fake_api.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    r = subprocess.run(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--run-mode", "scale-down"],
+        capture_output=True, text=True, timeout=60, env=env)
+    assert r.returncode == 0, r.stderr
+    assert fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner") is None
+    assert not any("/leases/" in p for (_, p) in fake_api.requests)
