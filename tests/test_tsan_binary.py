@@ -29,7 +29,8 @@ def tsan_bin():
 
 
 def test_tsan_daemon_watch_scaledown(tsan_bin, fake_api, fake_prom):
-    """Several daemon ticks with watch informers + OTLP under TSan."""
+    """Several daemon ticks with watch informers + OTLP + leader election
+    under TSan (elector thread included in the race surface)."""
     from gpu_pruner_amd.fixtures import FakeOtlpCollector, build_synthetic_cluster
 
     build_synthetic_cluster(fake_api, fake_prom, n_pods=40, pods_per_parent=2)
@@ -41,10 +42,13 @@ def test_tsan_daemon_watch_scaledown(tsan_bin, fake_api, fake_prom):
         env["OTEL_EXPORTER_OTLP_ENDPOINT"] = col.url
         env["OTEL_METRIC_EXPORT_INTERVAL"] = "500"
         env["TSAN_OPTIONS"] = "halt_on_error=0 exitcode=66"
+        env["POD_NAME"] = "tsan-replica"
+        env["POD_NAMESPACE"] = "gpu-pruner-system"
         p = subprocess.Popen(
             [tsan_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
              "--run-mode", "scale-down", "--check-interval", "1",
-             "--eval-strategy", "watch"],
+             "--eval-strategy", "watch", "--leader-elect",
+             "--leader-elect-renew-period", "1"],
             env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
         import time
         deadline = time.monotonic() + 30
