@@ -60,9 +60,12 @@ def test_two_replicas_single_leader_and_failover(pruner_bin, fake_api, fake_prom
     time.sleep(0.5)  # deterministic: a acquires first
     b = start_daemon(pruner_bin, fake_api, fake_prom, "replica-b")
     try:
-        # exactly one holder, and it culls
+        # exactly one holder, and it culls (on a loaded box either replica
+        # can win the initial race — pick winners/losers dynamically)
         holder = wait_for(lambda: lease_holder(fake_api), 10)
-        assert holder == "replica-a", holder
+        assert holder in ("replica-a", "replica-b"), holder
+        leader, standby = (a, b) if holder == "replica-a" else (b, a)
+        standby_id = "replica-b" if holder == "replica-a" else "replica-a"
         assert wait_for(
             lambda: fake_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0, 10)
         lease = fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner")
@@ -77,10 +80,10 @@ def test_two_replicas_single_leader_and_failover(pruner_bin, fake_api, fake_prom
 
         # hard-kill the leader (no release): the standby takes over after
         # the lease expires (4 s duration + renew cadence)
-        a.kill()
-        a.wait(timeout=10)
-        holder = wait_for(lambda: lease_holder(fake_api) == "replica-b", 15)
-        assert holder, f"standby never took over (holder={lease_holder(fake_api)})"
+        leader.kill()
+        leader.wait(timeout=10)
+        took_over = wait_for(lambda: lease_holder(fake_api) == standby_id, 20)
+        assert took_over, f"standby never took over (holder={lease_holder(fake_api)})"
         lease = fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner")
         assert lease["spec"]["leaseTransitions"] >= 1
 
@@ -90,10 +93,10 @@ def test_two_replicas_single_leader_and_failover(pruner_bin, fake_api, fake_prom
             lambda: fake_api.get("Deployment", "ml", "d")["spec"]["replicas"] == 0, 10)
 
         # graceful shutdown releases the lease immediately
-        b.send_signal(signal.SIGTERM)
-        b.wait(timeout=15)
+        standby.send_signal(signal.SIGTERM)
+        standby.wait(timeout=15)
         assert lease_holder(fake_api) == ""
-        err = b.stderr.read().decode()
+        err = standby.stderr.read().decode()
         assert "Acquired leadership" in err
         assert "Released lease" in err
     finally:
@@ -110,21 +113,23 @@ def test_standby_logs_and_does_not_act(pruner_bin, fake_api, fake_prom):
                      owner_uid=rs["metadata"]["uid"], age_s=3 * 3600)
     fake_prom.add_idle_series("p2", "ml")
 
-    a = start_daemon(pruner_bin, fake_api, fake_prom, "leader")
+    a = start_daemon(pruner_bin, fake_api, fake_prom, "first")
     time.sleep(0.5)
-    b = start_daemon(pruner_bin, fake_api, fake_prom, "standby")
+    b = start_daemon(pruner_bin, fake_api, fake_prom, "second")
     try:
-        assert wait_for(lambda: lease_holder(fake_api) == "leader", 10)
+        holder = wait_for(lambda: lease_holder(fake_api), 10)
+        assert holder in ("first", "second"), holder
+        leader_proc, standby_proc = (a, b) if holder == "first" else (b, a)
         time.sleep(2.5)
-        b.send_signal(signal.SIGTERM)
-        b.wait(timeout=15)
-        err = b.stderr.read().decode()
+        standby_proc.send_signal(signal.SIGTERM)
+        standby_proc.wait(timeout=15)
+        err = standby_proc.stderr.read().decode()
         assert "standing by" in err
         assert "Acquired leadership" not in err.replace(
             "Leadership acquired", "")  # never led
         # leader survives the standby's exit and still holds the lease
-        assert lease_holder(fake_api) == "leader"
-        assert a.poll() is None
+        assert lease_holder(fake_api) == holder
+        assert leader_proc.poll() is None
     finally:
         for p in (a, b):
             if p.poll() is None:
