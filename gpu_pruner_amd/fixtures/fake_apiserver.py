@@ -372,7 +372,13 @@ class FakeApiServer:
                     fixture._record_locked("ADDED", kind, obj)
                 return self._send(201, obj)
 
-        self._server = ThreadingHTTPServer((host, port), Handler)
+        # ThreadingHTTPServer's default listen backlog is 5: the engine's
+        # 32-way fan-out (plus informer reconnects) overflows it under
+        # load, surfacing as spurious connection-refused/reset
+        class _Server(ThreadingHTTPServer):
+            request_queue_size = 128
+
+        self._server = _Server((host, port), Handler)
         self._server.daemon_threads = True
         self._tls = certfile is not None
         if certfile is not None:

@@ -79,7 +79,13 @@ class FakePrometheus:
                 else:
                     self._send(404, {"status": "error", "error": "not found"})
 
-        self._server = ThreadingHTTPServer((host, port), Handler)
+        # ThreadingHTTPServer's default listen backlog is 5: the engine's
+        # 32-way fan-out (plus informer reconnects) overflows it under
+        # load, surfacing as spurious connection-refused/reset
+        class _Server(ThreadingHTTPServer):
+            request_queue_size = 128
+
+        self._server = _Server((host, port), Handler)
         self._server.daemon_threads = True
         self._tls = certfile is not None
         if certfile is not None:

@@ -67,7 +67,13 @@ class MiniProm:
                 q = urllib.parse.parse_qs(body).get("query", [""])[0]
                 self._query(q)
 
-        self._server = ThreadingHTTPServer((host, port), Handler)
+        # ThreadingHTTPServer's default listen backlog is 5: the engine's
+        # 32-way fan-out (plus informer reconnects) overflows it under
+        # load, surfacing as spurious connection-refused/reset
+        class _Server(ThreadingHTTPServer):
+            request_queue_size = 128
+
+        self._server = _Server((host, port), Handler)
         self._server.daemon_threads = True
         self._thread = threading.Thread(
             target=lambda: self._server.serve_forever(poll_interval=0.05), daemon=True)

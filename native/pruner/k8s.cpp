@@ -294,7 +294,23 @@ http::Response KubeClient::authed(const http::Request& req) {
   constexpr int kMaxRetries = 3;
   constexpr int kMaxWaitMs = 5000;
   for (int attempt = 0;; attempt++) {
-    http::Response resp = http_->request(r);
+    http::Response resp;
+    try {
+      resp = http_->request(r);
+    } catch (const http::Error& e) {
+      // Transient transport failure (RST/refused under load). For
+      // IDEMPOTENT verbs, retry briefly instead of surfacing: an
+      // owner-walk GET that errors makes the engine fall through to
+      // scaling the child (reference lib.rs:464 semantics) — correct as a
+      // last resort, but a 20 ms blip should not change which object gets
+      // scaled. Non-idempotent verbs (Event POST) surface immediately.
+      if (r.method != "GET" || attempt >= kMaxRetries) throw;
+      logx::counter_add("monotonic_counter.k8s_transport_retries", 1);
+      LOGD("pruner::k8s", std::string("transport error on GET ") + r.path + " (" +
+                              e.what() + "), retrying");
+      std::this_thread::sleep_for(std::chrono::milliseconds(20 * (attempt + 1)));
+      continue;
+    }
     if ((resp.status != 429 && resp.status != 503) || attempt >= kMaxRetries) return resp;
     logx::counter_add("monotonic_counter.k8s_throttled", 1);
     int wait_ms = 100 * (1 << attempt);  // backoff default when no header

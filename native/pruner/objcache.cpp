@@ -90,7 +90,15 @@ void ObjectCache::prefetch(const std::map<std::string, int>& ns_counts, int conc
 std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
                                                const std::string& name) {
   if (auto iit = informers_.find({ns, "Pod"}); iit != informers_.end()) {
-    if (iit->second->synced()) return iit->second->get(name);
+    if (iit->second->synced()) {
+      std::lock_guard<std::mutex> lock(memo_mu_);
+      auto key = std::make_tuple(std::string("Pod"), ns, name);
+      auto mit = memo_.find(key);
+      if (mit != memo_.end()) return mit->second;
+      auto v = iit->second->get(name);
+      memo_.emplace(std::move(key), v);
+      return v;
+    }
   }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     if (auto kit = nit->second.by_kind.find("Pod"); kit != nit->second.by_kind.end()) {
@@ -105,7 +113,16 @@ std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
 std::optional<jsn::Value> ObjectCache::get_object(Kind kind, const std::string& ns,
                                                   const std::string& name) {
   if (auto iit = informers_.find({ns, kind_name(kind)}); iit != informers_.end()) {
-    if (iit->second->synced()) return iit->second->get(name);
+    if (iit->second->synced()) {
+      // one snapshot per object per tick (see memo_ comment in the header)
+      std::lock_guard<std::mutex> lock(memo_mu_);
+      auto key = std::make_tuple(std::string(kind_name(kind)), ns, name);
+      auto mit = memo_.find(key);
+      if (mit != memo_.end()) return mit->second;
+      auto v = iit->second->get(name);
+      memo_.emplace(std::move(key), v);
+      return v;
+    }
   }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     auto kit = nit->second.by_kind.find(kind_name(kind));
