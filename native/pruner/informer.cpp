@@ -62,6 +62,11 @@ std::optional<jsn::Value> Informer::get(const std::string& name) const {
   return it->second;
 }
 
+std::map<std::string, jsn::Value> Informer::store_snapshot() const {
+  std::lock_guard<std::mutex> lock(mu_);
+  return store_;
+}
+
 size_t Informer::size() const {
   std::lock_guard<std::mutex> lock(mu_);
   return store_.size();

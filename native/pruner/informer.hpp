@@ -47,6 +47,11 @@ public:
 
   // nullopt when the object is not in the store (== gone, when synced).
   std::optional<jsn::Value> get(const std::string& name) const;
+  // Consistent copy of the whole store (values are COW-shared — ~one
+  // refcount bump per object). The per-tick snapshot every decision pass
+  // reads from: lookups inside a tick must all see ONE resourceVersion of
+  // each object or reference-parity struct-equality dedup double-counts.
+  std::map<std::string, jsn::Value> store_snapshot() const;
   size_t size() const;
   uint64_t lists_issued() const { return lists_issued_; }
   uint64_t events_seen() const { return events_seen_; }

@@ -36,12 +36,21 @@ void ObjectCache::prefetch(const std::map<std::string, int>& ns_counts, int conc
         informers_[{ns, kind_name(k)}] =
             &reg.get_or_create(kube_.config(), ns, kind_name(k), collection_path(k, ns));
     }
-    // wait for initial syncs (in parallel across informer threads); a
-    // not-yet-synced informer falls back to GETs this tick
-    for (auto& [key, inf] : informers_)
-      if (!inf->wait_synced(10000))
+    // wait for initial syncs (in parallel across informer threads), then
+    // SNAPSHOT each synced store into the per-tick cache: every lookup in
+    // this tick sees one consistent resourceVersion of each object (the
+    // informer keeps moving underneath — our own previous patches stream
+    // back as MODIFIED events — and reference-parity struct-equality dedup
+    // must not see two versions of one parent). A not-yet-synced informer
+    // stays uncached → GET fallback this tick.
+    for (auto& [key, inf] : informers_) {
+      if (!inf->wait_synced(10000)) {
         LOGW(TARGET, "informer for " + key.first + "/" + key.second +
                          " not synced; falling back to GETs this tick");
+        continue;
+      }
+      cache_[key.first].by_kind[key.second] = inf->store_snapshot();
+    }
     return;
   }
 
@@ -89,17 +98,6 @@ void ObjectCache::prefetch(const std::map<std::string, int>& ns_counts, int conc
 
 std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
                                                const std::string& name) {
-  if (auto iit = informers_.find({ns, "Pod"}); iit != informers_.end()) {
-    if (iit->second->synced()) {
-      std::lock_guard<std::mutex> lock(memo_mu_);
-      auto key = std::make_tuple(std::string("Pod"), ns, name);
-      auto mit = memo_.find(key);
-      if (mit != memo_.end()) return mit->second;
-      auto v = iit->second->get(name);
-      memo_.emplace(std::move(key), v);
-      return v;
-    }
-  }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     if (auto kit = nit->second.by_kind.find("Pod"); kit != nit->second.by_kind.end()) {
       auto oit = kit->second.find(name);
@@ -112,18 +110,6 @@ std::optional<jsn::Value> ObjectCache::get_pod(const std::string& ns,
 
 std::optional<jsn::Value> ObjectCache::get_object(Kind kind, const std::string& ns,
                                                   const std::string& name) {
-  if (auto iit = informers_.find({ns, kind_name(kind)}); iit != informers_.end()) {
-    if (iit->second->synced()) {
-      // one snapshot per object per tick (see memo_ comment in the header)
-      std::lock_guard<std::mutex> lock(memo_mu_);
-      auto key = std::make_tuple(std::string(kind_name(kind)), ns, name);
-      auto mit = memo_.find(key);
-      if (mit != memo_.end()) return mit->second;
-      auto v = iit->second->get(name);
-      memo_.emplace(std::move(key), v);
-      return v;
-    }
-  }
   if (auto nit = cache_.find(ns); nit != cache_.end()) {
     auto kit = nit->second.by_kind.find(kind_name(kind));
     if (kit != nit->second.by_kind.end()) {

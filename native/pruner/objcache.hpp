@@ -64,16 +64,6 @@ private:
   std::map<std::string, NsCache> cache_;  // namespace → cache
   // watch strategy: per-(ns, kind) informer handles for this tick's lookups
   std::map<std::pair<std::string, std::string>, Informer*> informers_;
-  // Per-tick snapshot memo (watch strategy): the informer store keeps
-  // moving (our own previous tick's patches stream back as MODIFIED
-  // events), but every lookup of the same object within ONE tick must see
-  // ONE snapshot — the reference's ScaleKind Eq is full struct equality
-  // for the builtin kinds (lib.rs:45-58), so two resourceVersions of the
-  // same Deployment would defeat the parent dedup and double-scale.
-  // (The LIST strategy gets this for free: a tick's LIST is one snapshot.)
-  std::mutex memo_mu_;
-  std::map<std::tuple<std::string, std::string, std::string>,
-           std::optional<jsn::Value>> memo_;
   size_t lists_issued_ = 0;
 };
 
