@@ -22,6 +22,7 @@
 #include "../common/http.hpp"
 #include "../common/miniyaml.hpp"
 #include "../pruner/informer.hpp"
+#include "../pruner/leader.hpp"
 #include "../pruner/otlp.hpp"
 #include "../pruner/synthbench.hpp"
 
@@ -280,6 +281,23 @@ PYBIND11_MODULE(_pruner_core, m) {
         },
         "Open a streaming GET and return every decoded body line "
         "(BodyStream chunk-decoder test surface)");
+
+  // Lease-based leader elector (k8s config from the environment) — lets
+  // tests drive single acquire/renew attempts deterministically.
+  py::class_<LeaderElector>(m, "LeaderElector")
+      .def(py::init([](const std::string& ns, const std::string& lease_name,
+                       const std::string& identity, int lease_duration_s,
+                       int renew_period_s) {
+             return new LeaderElector(KubeConfig::resolve(), ns, lease_name, identity,
+                                      lease_duration_s, renew_period_s);
+           }),
+           py::arg("namespace"), py::arg("lease_name"), py::arg("identity"),
+           py::arg("lease_duration_s") = 15, py::arg("renew_period_s") = 5)
+      .def("try_acquire_or_renew", &LeaderElector::try_acquire_or_renew,
+           py::call_guard<py::gil_scoped_release>())
+      .def("start", &LeaderElector::start)
+      .def("stop", &LeaderElector::stop, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("is_leader", &LeaderElector::is_leader);
 
   m.def("informers_reset",
         [] {

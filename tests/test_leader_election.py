@@ -153,3 +153,82 @@ def test_leader_elect_off_by_default(pruner_bin, fake_api, fake_prom):
     assert r.returncode == 0, r.stderr
     assert fake_api.get("Lease", "gpu-pruner-system", "gpu-pruner") is None
     assert not any("/leases/" in p for (_, p) in fake_api.requests)
+
+
+# ---- fine-grained elector semantics (pybind surface) ------------------------
+
+
+@pytest.fixture
+def elector_env(fake_api, monkeypatch):
+    monkeypatch.setenv("GPU_PRUNER_K8S_URL", fake_api.url)
+    return fake_api
+
+
+def test_elector_acquire_renew_and_conflict(core, elector_env):
+    api = elector_env
+    a = core.LeaderElector("ns1", "gpu-pruner", "id-a", 15, 5)
+    b = core.LeaderElector("ns1", "gpu-pruner", "id-b", 15, 5)
+    assert a.try_acquire_or_renew() is True
+    assert a.is_leader
+    # b cannot take a live lease
+    assert b.try_acquire_or_renew() is False
+    assert not b.is_leader
+    # a renews: renewTime advances, holder unchanged
+    lease1 = api.get("Lease", "ns1", "gpu-pruner")
+    assert a.try_acquire_or_renew() is True
+    lease2 = api.get("Lease", "ns1", "gpu-pruner")
+    assert lease2["spec"]["holderIdentity"] == "id-a"
+    assert lease2["spec"]["renewTime"] >= lease1["spec"]["renewTime"]
+    assert lease2["spec"]["leaseTransitions"] == 0
+
+
+def test_elector_takes_over_expired_lease(core, elector_env):
+    api = elector_env
+    # a stale lease: renewTime far in the past
+    api.put("Lease", {
+        "apiVersion": "coordination.k8s.io/v1", "kind": "Lease",
+        "metadata": {"name": "gpu-pruner", "namespace": "ns2"},
+        "spec": {"holderIdentity": "dead-replica", "leaseDurationSeconds": 15,
+                  "renewTime": "2020-01-01T00:00:00.000000Z",
+                  "acquireTime": "2020-01-01T00:00:00.000000Z",
+                  "leaseTransitions": 3},
+    })
+    b = core.LeaderElector("ns2", "gpu-pruner", "id-b", 15, 5)
+    assert b.try_acquire_or_renew() is True
+    lease = api.get("Lease", "ns2", "gpu-pruner")
+    assert lease["spec"]["holderIdentity"] == "id-b"
+    assert lease["spec"]["leaseTransitions"] == 4  # takeover counted
+
+
+def test_elector_takes_over_released_lease(core, elector_env):
+    api = elector_env
+    api.put("Lease", {
+        "apiVersion": "coordination.k8s.io/v1", "kind": "Lease",
+        "metadata": {"name": "gpu-pruner", "namespace": "ns3"},
+        "spec": {"holderIdentity": "", "leaseDurationSeconds": 15,
+                  "renewTime": "2020-01-01T00:00:00.000000Z",
+                  "leaseTransitions": 1},
+    })
+    b = core.LeaderElector("ns3", "gpu-pruner", "id-b", 15, 5)
+    assert b.try_acquire_or_renew() is True
+
+
+def test_elector_tolerates_malformed_lease(core, elector_env):
+    """A lease with a garbled/missing renewTime is treated as expired —
+    no crash, clean takeover."""
+    api = elector_env
+    api.put("Lease", {
+        "apiVersion": "coordination.k8s.io/v1", "kind": "Lease",
+        "metadata": {"name": "gpu-pruner", "namespace": "ns4"},
+        "spec": {"holderIdentity": "someone", "renewTime": "not-a-time"},
+    })
+    b = core.LeaderElector("ns4", "gpu-pruner", "id-b", 15, 5)
+    assert b.try_acquire_or_renew() is True
+    assert api.get("Lease", "ns4", "gpu-pruner")["spec"]["holderIdentity"] == "id-b"
+
+
+def test_elector_drops_leadership_when_apiserver_unreachable(core, monkeypatch):
+    monkeypatch.setenv("GPU_PRUNER_K8S_URL", "http://127.0.0.1:1")
+    e = core.LeaderElector("ns5", "gpu-pruner", "id-x", 15, 5)
+    assert e.try_acquire_or_renew() is False
+    assert not e.is_leader
