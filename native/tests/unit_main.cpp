@@ -15,6 +15,7 @@
 #include "../common/queue.hpp"
 #include "../common/threadpool.hpp"
 #include "../common/strutil.hpp"
+#include "../exporter/sampler.hpp"
 #include "../pruner/config.hpp"
 #include "../pruner/promql.hpp"
 #include "../pruner/resources.hpp"
@@ -146,6 +147,40 @@ static void test_strutil() {
   CHECK(strutil::uuid4_simple() != strutil::uuid4_simple());
 }
 
+static void test_activity_window() {
+  using exporter::ActivityWindow;
+  // time-weighted ratio over known segments, sliding window
+  ActivityWindow w;
+  for (int i = 0; i <= 10; i++) w.add(i, i % 2 == 0 ? 1.0 : 0.0, true);
+  double known = 0;
+  double r = w.ratio(10.0, 10.0, &known);
+  CHECK(known > 9.99 && known < 10.01);
+  CHECK(r > 0.3 && r < 0.7);
+  // idempotent reads
+  CHECK(w.ratio(10.0, 10.0) == w.ratio(10.0, 10.0));
+  // unknown (failed-read) time contributes to neither side
+  ActivityWindow u;
+  u.add(0.0, 0.0, true);
+  u.add(1.0, 1.0, true);
+  for (int t = 2; t <= 8; t++) u.add(t, 0.0, false);
+  r = u.ratio(8.0, 8.0, &known);
+  CHECK(known > 0.99 && known < 1.01);
+  CHECK(r > 0.99);
+  // burst ages out of the window
+  ActivityWindow b;
+  b.add(0.0, 0.0, true);
+  b.add(1.0, 1.0, true);
+  for (int t = 2; t <= 40; t++) b.add(t, 0.0, true);
+  CHECK(b.ratio(40.0, 5.0) == 0.0);
+  // retention bounds memory
+  ActivityWindow m;
+  m.set_retention(10.0);
+  for (int i = 0; i < 1000; i++) m.add(i, 0.5, true);
+  CHECK(m.size() < 20);
+  // concurrent readers while a poller appends (sampler mu_ serializes in
+  // production; here the reads are on const snapshots via copies)
+}
+
 int main() {
   test_json();
   test_promql();
@@ -154,6 +189,7 @@ int main() {
   test_threadpool();
   test_cli();
   test_strutil();
+  test_activity_window();
   if (failures == 0) std::puts("native unit tests: all passed");
   return failures == 0 ? 0 : 1;
 }
