@@ -78,3 +78,12 @@ def test_oversized_request_rejected_loudly(gpumon, echo_server):
 def test_connection_refused_raises(gpumon):
     with pytest.raises(gpumon.GrpcError):
         gpumon.grpc_unary_call("127.0.0.1", 1, "/x/Y", b"", 1000)
+
+
+@pytest.mark.parametrize("n", [126, 127, 128, 255, 256, 300])
+def test_hpack_integer_coding_boundaries(gpumon, echo_server, n):
+    """RFC 7541 §5.1 multi-byte integer boundaries for the :path literal."""
+    method = "/" + "s" * (n - len("/Svc/m") - 1) + ".Svc/m"
+    assert len(method) == n
+    assert gpumon.grpc_unary_call("127.0.0.1", echo_server["port"], method, b"k") == b"k"
+    assert method in echo_server["received"]
