@@ -48,7 +48,8 @@ def test_transcripts_cover_every_pruner_verb():
     POST, 401 and 429 — so fake-fidelity claims cover the whole surface."""
     seen = {"get_200": False, "get_404": False, "list": False,
             "merge_patch": False, "scale": False, "event_post": False,
-            "unauthorized": False, "throttle": False}
+            "unauthorized": False, "throttle": False,
+            "lease_put": False, "conflict": False}
     for path in TRANSCRIPTS:
         t = json.loads(path.read_text())
         for step in t["steps"]:
@@ -70,5 +71,9 @@ def test_transcripts_cover_every_pruner_verb():
                 seen["unauthorized"] = True
             if s == 429:
                 seen["throttle"] = True
+            if m == "PUT" and "/leases/" in p and s in (200, 409):
+                seen["lease_put"] = True
+            if s == 409:
+                seen["conflict"] = True
     missing = [k for k, v in seen.items() if not v]
     assert not missing, f"conformance gaps: {missing}"
