@@ -143,7 +143,20 @@ int main(int argc, char** argv) {
       resp.body = exporter::render_metrics(samples, attribs, opts);
       resp.content_type = "text/plain; version=0.0.4; charset=utf-8";
     } else if (req.path == "/healthz") {
-      resp.body = "ok\n";
+      // liveness reflects the sampler: if EVERY device's activity reads are
+      // failing, the exporter is not doing its job — let the kubelet
+      // restart it (per-device health is also exported as
+      // mi355_sampler_healthy for alerting)
+      auto samples = sampler.snapshot();
+      bool any_healthy = false;
+      for (const auto& d : samples)
+        if (d.healthy) any_healthy = true;
+      if (any_healthy) {
+        resp.body = "ok\n";
+      } else {
+        resp.status = 503;
+        resp.body = "no device with working activity reads\n";
+      }
     } else {
       resp.status = 404;
       resp.body = "not found; try /metrics\n";
