@@ -181,6 +181,9 @@ PYBIND11_MODULE(_pruner_core, m) {
                                                                      : "default";
     c["max_concurrency"] = r.config.max_concurrency;
     c["queue_capacity"] = r.config.queue_capacity;
+    c["leader_elect"] = r.config.leader_elect;
+    c["leader_lease_duration_s"] = r.config.leader_lease_duration_s;
+    c["leader_renew_period_s"] = r.config.leader_renew_period_s;
     d["config"] = c;
     return d;
   });

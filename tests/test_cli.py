@@ -109,3 +109,32 @@ def test_invalid_log_format_rejected(core):
 def test_help(core):
     r = parse(core, "--help")
     assert r["help"] is True
+
+
+def test_leader_elect_flags(core):
+    r = parse(core, "--prometheus-url", "http://p", "--leader-elect",
+              "--leader-elect-lease-duration", "30",
+              "--leader-elect-renew-period", "10")
+    assert r["config"]["leader_elect"] is True
+    assert r["config"]["leader_lease_duration_s"] == 30
+    assert r["config"]["leader_renew_period_s"] == 10
+
+
+def test_leader_elect_defaults(core):
+    r = parse(core, "--prometheus-url", "http://p")
+    assert r["config"]["leader_elect"] is False
+    assert r["config"]["leader_lease_duration_s"] == 15
+    assert r["config"]["leader_renew_period_s"] == 5
+
+
+def test_help_text_covers_new_flags():
+    import subprocess
+    from pathlib import Path
+
+    binary = Path(__file__).resolve().parent.parent / "bin" / "gpu-pruner"
+    r = subprocess.run([str(binary), "--help"], capture_output=True, text=True,
+                       timeout=30)
+    assert r.returncode == 0
+    for flag in ("--leader-elect", "--eval-strategy", "--metrics-port",
+                 "--max-concurrency", "--honor-labels", "--power-threshold"):
+        assert flag in r.stdout, flag
