@@ -45,7 +45,8 @@ def test_fake_apiserver_conformance(path):
 def test_transcripts_cover_every_pruner_verb():
     """The suite must exercise every apiserver interaction the pruner
     performs: GET (found + 404), LIST, merge-PATCH, /scale PATCH, Event
-    POST, 401 and 429 — so fake-fidelity claims cover the whole surface."""
+    POST, Lease PUT with optimistic concurrency (200 + 409), 401 and 429 —
+    so fake-fidelity claims cover the whole surface."""
     seen = {"get_200": False, "get_404": False, "list": False,
             "merge_patch": False, "scale": False, "event_post": False,
             "unauthorized": False, "throttle": False,
