@@ -46,9 +46,49 @@ def test_clusterrole_covers_daemon_surface():
         ("apps", "statefulsets"), ("apps", "statefulsets/scale"),
         ("kubeflow.org", "notebooks"),
         ("serving.kserve.io", "inferenceservices"),
+        ("coordination.k8s.io", "leases"),  # --leader-elect
     ]
     for pair in needed:
         assert pair in granted, f"ClusterRole missing {pair}"
+
+
+def test_clusterrole_lease_verbs():
+    """--leader-elect needs exactly get/create/update on Leases: get to read
+    the current holder, create for first acquisition, update (rv-fenced PUT)
+    for renew/takeover. list/watch/delete are NOT used by the elector and
+    must not be granted (least privilege)."""
+    docs = load_all("clusterrole.yaml")
+    pruner_cr = next(d for d in docs if d["metadata"]["name"] == "gpu-pruner-cr")
+    lease_rules = [r for r in pruner_cr["rules"]
+                   if "coordination.k8s.io" in r["apiGroups"]
+                   and "leases" in r["resources"]]
+    assert lease_rules, "no Lease rule in gpu-pruner-cr"
+    verbs = {v for r in lease_rules for v in r["verbs"]}
+    assert verbs == {"get", "create", "update"}, verbs
+
+
+def test_crd_manifests_parse_as_v1_crds():
+    """deploy/crds/ ships minimal Notebook + InferenceService CRDs so the
+    envtest/e2e tiers (and air-gapped installs without Kubeflow/KServe) can
+    register the custom kinds the walk resolves. Both must be well-formed
+    apiextensions.k8s.io/v1 with the group/kind the engine expects."""
+    crd_dir = DEPLOY / "crds"
+    expected = {
+        ("kubeflow.org", "Notebook"),
+        ("serving.kserve.io", "InferenceService"),
+    }
+    seen = set()
+    for f in sorted(crd_dir.glob("*.yaml")):
+        for d in yaml.safe_load_all(f.read_text()):
+            assert d["apiVersion"] == "apiextensions.k8s.io/v1", f.name
+            assert d["kind"] == "CustomResourceDefinition", f.name
+            spec = d["spec"]
+            names = spec["names"]
+            seen.add((spec["group"], names["kind"]))
+            # name must be <plural>.<group> per apiserver validation
+            assert d["metadata"]["name"] == f"{names['plural']}.{spec['group']}"
+            assert any(v.get("served") for v in spec["versions"]), f.name
+    assert seen == expected, seen
 
 
 def test_deployment_resource_budget():
