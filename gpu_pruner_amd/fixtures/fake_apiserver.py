@@ -79,6 +79,10 @@ class FakeApiServer:
         self.watch_requests = 0  # watch connections served (test observability)
         # failure injection for informer resilience tests
         self.watch_410_next = 0  # respond 410 Gone to the next N watch requests
+        # conflict injection: respond 409 to the next N PUTs regardless of
+        # resourceVersion — simulates another writer winning the GET→PUT
+        # race window (leader-election renew conflicts)
+        self.conflict_next_put = 0
 
         fixture = self
 
@@ -312,6 +316,14 @@ class FakeApiServer:
                                                         "the server could not find the requested resource"))
                 kind, ns, name, _sub = r
                 with fixture._lock:
+                    if fixture.conflict_next_put > 0:
+                        fixture.conflict_next_put -= 1
+                        return self._send(409, self._status(
+                            409, "Conflict",
+                            f'Operation cannot be fulfilled on {kind.lower()}s "{name}": '
+                            "the object has been modified; please apply your changes to "
+                            "the latest version and try again",
+                            details={"name": name, "kind": kind.lower() + "s"}))
                     obj = fixture.objects.get((kind, ns, name))
                     if obj is None:
                         return self._send(404, self._status(

@@ -85,8 +85,21 @@ bool LeaderElector::try_acquire_or_renew() {
     bool expired = !have_renew || now_unix() > renew_s + static_cast<double>(duration);
 
     if (holder == identity_) {
-      // renew our own lease
-      kube.replace(lease_path(), lease_body(identity_, rv, acquire_time, transitions));
+      // renew our own lease (rv-fenced: someone force-updating or deleting
+      // the Lease between our GET and this PUT means our claim is stale —
+      // treat it as lost and re-contest next tick, don't keep acting)
+      try {
+        kube.replace(lease_path(), lease_body(identity_, rv, acquire_time, transitions));
+      } catch (const KubeError& e) {
+        if (e.status == 409 || e.status == 404) {
+          if (leading_.exchange(false))
+            LOGW(TARGET, "Lost leadership of " + ns_ + "/" + name_ + " (lease " +
+                             (e.status == 409 ? "conflict" : "deleted") +
+                             " during renew)");
+          return false;
+        }
+        throw;
+      }
       if (!leading_.exchange(true))
         LOGI(TARGET, "Re-acquired leadership of " + ns_ + "/" + name_);
       return true;

@@ -227,6 +227,36 @@ def test_elector_tolerates_malformed_lease(core, elector_env):
     assert api.get("Lease", "ns4", "gpu-pruner")["spec"]["holderIdentity"] == "id-b"
 
 
+def test_elector_loses_leadership_on_renew_conflict(core, elector_env):
+    """A 409 on renewing our OWN lease (another writer won the GET→PUT race
+    window — forced update, parallel controller) must drop leadership and
+    re-contest next tick, not abort or keep acting on a stale claim."""
+    api = elector_env
+    e = core.LeaderElector("ns6", "gpu-pruner", "id-a", 15, 5)
+    assert e.try_acquire_or_renew() is True
+    api.conflict_next_put = 1  # next PUT → 409 regardless of resourceVersion
+    assert e.try_acquire_or_renew() is False
+    assert not e.is_leader
+    # injection consumed: the following tick re-contests and wins again
+    # (holder is still id-a in the store, so the renew path succeeds)
+    assert e.try_acquire_or_renew() is True
+    assert e.is_leader
+
+
+def test_elector_survives_lease_deleted_underneath(core, elector_env):
+    """Deleting the Lease while we hold it: the next tick's GET sees no
+    lease and the elector recreates it (staying leader) — the renew PUT
+    racing a delete (404) is the drop-and-recontest path."""
+    api = elector_env
+    e = core.LeaderElector("ns7", "gpu-pruner", "id-a", 15, 5)
+    assert e.try_acquire_or_renew() is True
+    api.delete_object("Lease", "ns7", "gpu-pruner")
+    assert api.get("Lease", "ns7", "gpu-pruner") is None
+    assert e.try_acquire_or_renew() is True  # recreated
+    lease = api.get("Lease", "ns7", "gpu-pruner")
+    assert lease["spec"]["holderIdentity"] == "id-a"
+
+
 def test_elector_drops_leadership_when_apiserver_unreachable(core, monkeypatch):
     monkeypatch.setenv("GPU_PRUNER_K8S_URL", "http://127.0.0.1:1")
     e = core.LeaderElector("ns5", "gpu-pruner", "id-x", 15, 5)
