@@ -209,7 +209,7 @@ void Server::handle_conn(int fd) {
         std::string chunk;
         more = resp.streamer(&chunk);
         if (!chunk.empty()) {
-          char sz[16];
+          char sz[24];  // 16 hex digits (64-bit size_t) + CRLF + NUL
           std::snprintf(sz, sizeof sz, "%zx\r\n", chunk.size());
           if (!send_all(std::string(sz) + chunk + "\r\n")) {
             ::close(fd);

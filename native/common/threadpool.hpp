@@ -118,11 +118,11 @@ private:
     if (cpus <= 0) {  // v1 pair
       long long q = -1, p = -1;
       if (FILE* f = std::fopen("/sys/fs/cgroup/cpu/cpu.cfs_quota_us", "r")) {
-        std::fscanf(f, "%lld", &q);
+        if (std::fscanf(f, "%lld", &q) != 1) q = -1;
         std::fclose(f);
       }
       if (FILE* f = std::fopen("/sys/fs/cgroup/cpu/cpu.cfs_period_us", "r")) {
-        std::fscanf(f, "%lld", &p);
+        if (std::fscanf(f, "%lld", &p) != 1) p = -1;
         std::fclose(f);
       }
       if (q > 0 && p > 0) cpus = static_cast<double>(q) / p;
