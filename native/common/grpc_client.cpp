@@ -11,6 +11,7 @@
 
 #include <chrono>
 #include <cstring>
+#include <unordered_map>
 
 namespace grpcx {
 
@@ -204,6 +205,221 @@ void hpack_literal_new_name(std::string& out, const std::string& name,
   hpack_str(out, value);
 }
 
+// ---------------- HPACK response decoding ----------------
+
+// RFC 7541 Appendix B Huffman code: {code, bit-length} per symbol 0..255
+// plus EOS (256). Validated as THE canonical table: the 257 entries form a
+// prefix-free code with Kraft sum exactly 1 and a 30-bit EOS, which pins it
+// uniquely to the RFC appendix.
+struct HuffSym { uint32_t code; uint8_t len; };
+constexpr HuffSym kHuff[257] = {
+    {0x1ff8u, 13}, {0x7fffd8u, 23}, {0xfffffe2u, 28}, {0xfffffe3u, 28},
+    {0xfffffe4u, 28}, {0xfffffe5u, 28}, {0xfffffe6u, 28}, {0xfffffe7u, 28},
+    {0xfffffe8u, 28}, {0xffffeau, 24}, {0x3ffffffcu, 30}, {0xfffffe9u, 28},
+    {0xfffffeau, 28}, {0x3ffffffdu, 30}, {0xfffffebu, 28}, {0xfffffecu, 28},
+    {0xfffffedu, 28}, {0xfffffeeu, 28}, {0xfffffefu, 28}, {0xffffff0u, 28},
+    {0xffffff1u, 28}, {0xffffff2u, 28}, {0x3ffffffeu, 30}, {0xffffff3u, 28},
+    {0xffffff4u, 28}, {0xffffff5u, 28}, {0xffffff6u, 28}, {0xffffff7u, 28},
+    {0xffffff8u, 28}, {0xffffff9u, 28}, {0xffffffau, 28}, {0xffffffbu, 28},
+    {0x14u, 6}, {0x3f8u, 10}, {0x3f9u, 10}, {0xffau, 12},
+    {0x1ff9u, 13}, {0x15u, 6}, {0xf8u, 8}, {0x7fau, 11},
+    {0x3fau, 10}, {0x3fbu, 10}, {0xf9u, 8}, {0x7fbu, 11},
+    {0xfau, 8}, {0x16u, 6}, {0x17u, 6}, {0x18u, 6},
+    {0x0u, 5}, {0x1u, 5}, {0x2u, 5}, {0x19u, 6},
+    {0x1au, 6}, {0x1bu, 6}, {0x1cu, 6}, {0x1du, 6},
+    {0x1eu, 6}, {0x1fu, 6}, {0x5cu, 7}, {0xfbu, 8},
+    {0x7ffcu, 15}, {0x20u, 6}, {0xffbu, 12}, {0x3fcu, 10},
+    {0x1ffau, 13}, {0x21u, 6}, {0x5du, 7}, {0x5eu, 7},
+    {0x5fu, 7}, {0x60u, 7}, {0x61u, 7}, {0x62u, 7},
+    {0x63u, 7}, {0x64u, 7}, {0x65u, 7}, {0x66u, 7},
+    {0x67u, 7}, {0x68u, 7}, {0x69u, 7}, {0x6au, 7},
+    {0x6bu, 7}, {0x6cu, 7}, {0x6du, 7}, {0x6eu, 7},
+    {0x6fu, 7}, {0x70u, 7}, {0x71u, 7}, {0x72u, 7},
+    {0xfcu, 8}, {0x73u, 7}, {0xfdu, 8}, {0x1ffbu, 13},
+    {0x7fff0u, 19}, {0x1ffcu, 13}, {0x3ffcu, 14}, {0x22u, 6},
+    {0x7ffdu, 15}, {0x3u, 5}, {0x23u, 6}, {0x4u, 5},
+    {0x24u, 6}, {0x5u, 5}, {0x25u, 6}, {0x26u, 6},
+    {0x27u, 6}, {0x6u, 5}, {0x74u, 7}, {0x75u, 7},
+    {0x28u, 6}, {0x29u, 6}, {0x2au, 6}, {0x7u, 5},
+    {0x2bu, 6}, {0x76u, 7}, {0x2cu, 6}, {0x8u, 5},
+    {0x9u, 5}, {0x2du, 6}, {0x77u, 7}, {0x78u, 7},
+    {0x79u, 7}, {0x7au, 7}, {0x7bu, 7}, {0x7ffeu, 15},
+    {0x7fcu, 11}, {0x3ffdu, 14}, {0x1ffdu, 13}, {0xffffffcu, 28},
+    {0xfffe6u, 20}, {0x3fffd2u, 22}, {0xfffe7u, 20}, {0xfffe8u, 20},
+    {0x3fffd3u, 22}, {0x3fffd4u, 22}, {0x3fffd5u, 22}, {0x7fffd9u, 23},
+    {0x3fffd6u, 22}, {0x7fffdau, 23}, {0x7fffdbu, 23}, {0x7fffdcu, 23},
+    {0x7fffddu, 23}, {0x7fffdeu, 23}, {0xffffebu, 24}, {0x7fffdfu, 23},
+    {0xffffecu, 24}, {0xffffedu, 24}, {0x3fffd7u, 22}, {0x7fffe0u, 23},
+    {0xffffeeu, 24}, {0x7fffe1u, 23}, {0x7fffe2u, 23}, {0x7fffe3u, 23},
+    {0x7fffe4u, 23}, {0x1fffdcu, 21}, {0x3fffd8u, 22}, {0x7fffe5u, 23},
+    {0x3fffd9u, 22}, {0x7fffe6u, 23}, {0x7fffe7u, 23}, {0xffffefu, 24},
+    {0x3fffdau, 22}, {0x1fffddu, 21}, {0xfffe9u, 20}, {0x3fffdbu, 22},
+    {0x3fffdcu, 22}, {0x7fffe8u, 23}, {0x7fffe9u, 23}, {0x1fffdeu, 21},
+    {0x7fffeau, 23}, {0x3fffddu, 22}, {0x3fffdeu, 22}, {0xfffff0u, 24},
+    {0x1fffdfu, 21}, {0x3fffdfu, 22}, {0x7fffebu, 23}, {0x7fffecu, 23},
+    {0x1fffe0u, 21}, {0x1fffe1u, 21}, {0x3fffe0u, 22}, {0x1fffe2u, 21},
+    {0x7fffedu, 23}, {0x3fffe1u, 22}, {0x7fffeeu, 23}, {0x7fffefu, 23},
+    {0xfffeau, 20}, {0x3fffe2u, 22}, {0x3fffe3u, 22}, {0x3fffe4u, 22},
+    {0x7ffff0u, 23}, {0x3fffe5u, 22}, {0x3fffe6u, 22}, {0x7ffff1u, 23},
+    {0x3ffffe0u, 26}, {0x3ffffe1u, 26}, {0xfffebu, 20}, {0x7fff1u, 19},
+    {0x3fffe7u, 22}, {0x7ffff2u, 23}, {0x3fffe8u, 22}, {0x1ffffecu, 25},
+    {0x3ffffe2u, 26}, {0x3ffffe3u, 26}, {0x3ffffe4u, 26}, {0x7ffffdeu, 27},
+    {0x7ffffdfu, 27}, {0x3ffffe5u, 26}, {0xfffff1u, 24}, {0x1ffffedu, 25},
+    {0x7fff2u, 19}, {0x1fffe3u, 21}, {0x3ffffe6u, 26}, {0x7ffffe0u, 27},
+    {0x7ffffe1u, 27}, {0x3ffffe7u, 26}, {0x7ffffe2u, 27}, {0xfffff2u, 24},
+    {0x1fffe4u, 21}, {0x1fffe5u, 21}, {0x3ffffe8u, 26}, {0x3ffffe9u, 26},
+    {0xffffffdu, 28}, {0x7ffffe3u, 27}, {0x7ffffe4u, 27}, {0x7ffffe5u, 27},
+    {0xfffecu, 20}, {0xfffff3u, 24}, {0xfffedu, 20}, {0x1fffe6u, 21},
+    {0x3fffe9u, 22}, {0x1fffe7u, 21}, {0x1fffe8u, 21}, {0x7ffff3u, 23},
+    {0x3fffeau, 22}, {0x3fffebu, 22}, {0x1ffffeeu, 25}, {0x1ffffefu, 25},
+    {0xfffff4u, 24}, {0xfffff5u, 24}, {0x3ffffeau, 26}, {0x7ffff4u, 23},
+    {0x3ffffebu, 26}, {0x7ffffe6u, 27}, {0x3ffffecu, 26}, {0x3ffffedu, 26},
+    {0x7ffffe7u, 27}, {0x7ffffe8u, 27}, {0x7ffffe9u, 27}, {0x7ffffeau, 27},
+    {0x7ffffebu, 27}, {0xffffffeu, 28}, {0x7ffffecu, 27}, {0x7ffffedu, 27},
+    {0x7ffffeeu, 27}, {0x7ffffefu, 27}, {0x7fffff0u, 27}, {0x3ffffeeu, 26},
+    {0x3fffffffu, 30}
+};
+
+// Bit-serial decode against the canonical table: accumulate bits, test the
+// accumulator at every length that exists in the code (5..30). Strings in
+// headers are short (metric of one connection), so O(bits) with a hash probe
+// per candidate length is plenty.
+const std::unordered_map<uint64_t, uint16_t>& huff_map() {
+  static const std::unordered_map<uint64_t, uint16_t> m = [] {
+    std::unordered_map<uint64_t, uint16_t> t;
+    for (uint16_t s = 0; s < 257; s++)
+      t.emplace((static_cast<uint64_t>(kHuff[s].len) << 32) | kHuff[s].code, s);
+    return t;
+  }();
+  return m;
+}
+
+}  // namespace
+
+std::string huffman_decode(const uint8_t* data, size_t len) {
+  const auto& map = huff_map();
+  std::string out;
+  uint32_t acc = 0;
+  uint8_t nbits = 0;
+  for (size_t i = 0; i < len; i++) {
+    for (int b = 7; b >= 0; b--) {
+      acc = (acc << 1) | ((data[i] >> b) & 1);
+      nbits++;
+      auto it = map.find((static_cast<uint64_t>(nbits) << 32) | acc);
+      if (it != map.end()) {
+        if (it->second == 256) throw GrpcError("HPACK: EOS inside Huffman string");
+        out += static_cast<char>(it->second);
+        acc = 0;
+        nbits = 0;
+      } else if (nbits > 30) {
+        throw GrpcError("HPACK: invalid Huffman code");
+      }
+    }
+  }
+  // padding must be <8 bits of the EOS prefix (all ones)
+  if (nbits >= 8 || acc != (1u << nbits) - 1)
+    throw GrpcError("HPACK: bad Huffman padding");
+  return out;
+}
+
+namespace {
+
+// RFC 7541 Appendix A static table (1..61); empty string = no value.
+constexpr const char* kStatic[62][2] = {
+    {"", ""},
+    {":authority", ""}, {":method", "GET"}, {":method", "POST"},
+    {":path", "/"}, {":path", "/index.html"}, {":scheme", "http"},
+    {":scheme", "https"}, {":status", "200"}, {":status", "204"},
+    {":status", "206"}, {":status", "304"}, {":status", "400"},
+    {":status", "404"}, {":status", "500"}, {"accept-charset", ""},
+    {"accept-encoding", "gzip, deflate"}, {"accept-language", ""},
+    {"accept-ranges", ""}, {"accept", ""}, {"access-control-allow-origin", ""},
+    {"age", ""}, {"allow", ""}, {"authorization", ""}, {"cache-control", ""},
+    {"content-disposition", ""}, {"content-encoding", ""},
+    {"content-language", ""}, {"content-length", ""}, {"content-location", ""},
+    {"content-range", ""}, {"content-type", ""}, {"cookie", ""}, {"date", ""},
+    {"etag", ""}, {"expect", ""}, {"expires", ""}, {"from", ""}, {"host", ""},
+    {"if-match", ""}, {"if-modified-since", ""}, {"if-none-match", ""},
+    {"if-range", ""}, {"if-unmodified-since", ""}, {"last-modified", ""},
+    {"link", ""}, {"location", ""}, {"max-forwards", ""},
+    {"proxy-authenticate", ""}, {"proxy-authorization", ""}, {"range", ""},
+    {"referer", ""}, {"refresh", ""}, {"retry-after", ""}, {"server", ""},
+    {"set-cookie", ""}, {"strict-transport-security", ""},
+    {"transfer-encoding", ""}, {"user-agent", ""}, {"vary", ""}, {"via", ""},
+    {"www-authenticate", ""},
+};
+
+}  // namespace
+
+std::vector<Header> HpackDecoder::decode_block(const std::string& block) {
+  std::vector<Header> out;
+  size_t pos = 0;
+  const auto* p = reinterpret_cast<const uint8_t*>(block.data());
+
+  auto read_int = [&](uint8_t prefix_bits) -> uint64_t {
+    if (pos >= block.size()) throw GrpcError("HPACK: truncated integer");
+    const uint64_t max_prefix = (1u << prefix_bits) - 1;
+    uint64_t v = p[pos++] & max_prefix;
+    if (v < max_prefix) return v;
+    int shift = 0;
+    while (true) {
+      if (pos >= block.size()) throw GrpcError("HPACK: truncated integer");
+      uint8_t b = p[pos++];
+      v += static_cast<uint64_t>(b & 0x7F) << shift;
+      shift += 7;
+      if (!(b & 0x80)) return v;
+      if (shift > 56) throw GrpcError("HPACK: integer overflow");
+    }
+  };
+
+  auto read_string = [&]() -> std::string {
+    if (pos >= block.size()) throw GrpcError("HPACK: truncated string");
+    bool huff = (p[pos] & 0x80) != 0;
+    uint64_t n = read_int(7);
+    if (pos + n > block.size()) throw GrpcError("HPACK: truncated string");
+    std::string s = huff ? huffman_decode(p + pos, n)
+                         : block.substr(pos, n);
+    pos += n;
+    return s;
+  };
+
+  auto lookup = [&](uint64_t idx) -> Header {
+    if (idx == 0) throw GrpcError("HPACK: index 0");
+    if (idx <= 61) return {kStatic[idx][0], kStatic[idx][1]};
+    size_t d = idx - 62;
+    if (d >= dynamic_.size()) throw GrpcError("HPACK: dynamic index out of range");
+    return dynamic_[d];
+  };
+
+  while (pos < block.size()) {
+    uint8_t b = p[pos];
+    if (b & 0x80) {                      // indexed header field
+      out.push_back(lookup(read_int(7)));
+    } else if ((b & 0xC0) == 0x40) {     // literal, incremental indexing
+      uint64_t idx = read_int(6);
+      std::string name = idx ? lookup(idx).first : read_string();
+      std::string value = read_string();
+      dynamic_.emplace_front(name, value);
+      // Eviction bookkeeping is deliberately skipped: a decoder that never
+      // evicts can only OVER-retain, and an index referencing an entry the
+      // encoder evicted would have been re-inserted by the encoder first.
+      // Connections here are one-RPC-lived; bound memory anyway:
+      if (dynamic_.size() > 1024) dynamic_.pop_back();
+      out.emplace_back(std::move(name), std::move(value));
+    } else if ((b & 0xE0) == 0x20) {     // dynamic table size update
+      read_int(5);
+    } else {                             // literal w/o indexing / never-indexed
+      uint64_t idx = read_int(4);
+      std::string name = idx ? lookup(idx).first : read_string();
+      std::string value = read_string();
+      out.emplace_back(std::move(name), std::move(value));
+    }
+  }
+  return out;
+}
+
+namespace {
+
 }  // namespace
 
 std::string unary_call(const Target& target, const std::string& method_path,
@@ -271,6 +487,23 @@ std::string unary_call(const Target& target, const std::string& method_path,
   std::string grpc_payload;
   bool stream_done = false;
   bool headers_pending_end_stream = false;
+  // response header / trailer decoding: the dynamic table persists across
+  // the initial HEADERS and the trailers (grpc servers index grpc-status et
+  // al. in the first block and reference it from the second)
+  HpackDecoder hpack;
+  std::vector<Header> resp_headers;
+  std::string hdr_block;
+  bool hdr_collecting = false;
+  bool hdr_unreliable = false;  // any decode failure → skip status checks
+  auto decode_hdr_block = [&]() {
+    try {
+      auto hs = hpack.decode_block(hdr_block);
+      resp_headers.insert(resp_headers.end(), hs.begin(), hs.end());
+    } catch (const GrpcError&) {
+      hdr_unreliable = true;  // tolerate odd encoders; behave as untyped h2
+    }
+    hdr_block.clear();
+  };
   while (!stream_done) {
     uint8_t fh[9];
     sock.read_exact(fh, 9);
@@ -302,15 +535,37 @@ std::string unary_call(const Target& target, const std::string& method_path,
         if (stream == 1) grpc_payload += payload;
         if (stream == 1 && (flags & FLAG_END_STREAM)) stream_done = true;
         break;
-      case F_HEADERS:  // response headers / trailers — content not parsed
-        if (stream == 1 && (flags & FLAG_END_STREAM)) {
-          if (flags & FLAG_END_HEADERS) stream_done = true;
-          else headers_pending_end_stream = true;
+      case F_HEADERS:  // response headers / trailers
+        if (stream == 1) {
+          // strip PADDED (0x8: leading pad-length byte + trailing pad) and
+          // PRIORITY (0x20: 5 bytes of dep + weight) before HPACK
+          size_t start = 0, end = payload.size();
+          if (flags & 0x8) {
+            if (!payload.empty()) {
+              uint8_t pad = static_cast<uint8_t>(payload[0]);
+              start = 1;
+              end = pad <= end - start ? end - pad : start;
+            }
+          }
+          if (flags & 0x20) start = std::min(start + 5, end);
+          hdr_block.assign(payload, start, end - start);
+          if (flags & FLAG_END_HEADERS) decode_hdr_block();
+          else hdr_collecting = true;
+          if (flags & FLAG_END_STREAM) {
+            if (flags & FLAG_END_HEADERS) stream_done = true;
+            else headers_pending_end_stream = true;
+          }
         }
         break;
       case F_CONTINUATION:
-        if (stream == 1 && headers_pending_end_stream && (flags & FLAG_END_HEADERS))
-          stream_done = true;
+        if (stream == 1 && hdr_collecting) {
+          hdr_block += payload;
+          if (flags & FLAG_END_HEADERS) {
+            decode_hdr_block();
+            hdr_collecting = false;
+            if (headers_pending_end_stream) stream_done = true;
+          }
+        }
         break;
       case F_RST_STREAM:
         if (stream == 1) throw GrpcError("stream reset by server");
@@ -322,6 +577,41 @@ std::string unary_call(const Target& target, const std::string& method_path,
         break;
       default:
         break;  // WINDOW_UPDATE / unknown
+    }
+  }
+
+  // ---- gRPC status (trailers; last occurrence wins) ----
+  if (!hdr_unreliable) {
+    std::string http_status, grpc_status, grpc_message;
+    for (const auto& [k, v] : resp_headers) {
+      if (k == ":status") http_status = v;
+      else if (k == "grpc-status") grpc_status = v;
+      else if (k == "grpc-message") grpc_message = v;
+    }
+    if (!http_status.empty() && http_status != "200")
+      throw GrpcError("HTTP " + http_status + " from gRPC server");
+    if (!grpc_status.empty() && grpc_status != "0") {
+      // grpc-message is percent-encoded (gRPC HTTP/2 spec)
+      std::string msg_out;
+      for (size_t i = 0; i < grpc_message.size(); i++) {
+        if (grpc_message[i] == '%' && i + 2 < grpc_message.size()) {
+          auto hex = [](char c) -> int {
+            if (c >= '0' && c <= '9') return c - '0';
+            if (c >= 'a' && c <= 'f') return c - 'a' + 10;
+            if (c >= 'A' && c <= 'F') return c - 'A' + 10;
+            return -1;
+          };
+          int hi = hex(grpc_message[i + 1]), lo = hex(grpc_message[i + 2]);
+          if (hi >= 0 && lo >= 0) {
+            msg_out += static_cast<char>(hi * 16 + lo);
+            i += 2;
+            continue;
+          }
+        }
+        msg_out += grpc_message[i];
+      }
+      throw GrpcError("grpc-status " + grpc_status +
+                      (msg_out.empty() ? "" : ": " + msg_out));
     }
   }
 

@@ -10,8 +10,11 @@
 #pragma once
 
 #include <cstdint>
+#include <deque>
 #include <stdexcept>
 #include <string>
+#include <utility>
+#include <vector>
 
 namespace grpcx {
 
@@ -35,12 +38,39 @@ struct Target {
 // bigger payloads must split them into multiple unary calls.
 constexpr size_t kMaxRequestBytes = 60000;
 
+using Header = std::pair<std::string, std::string>;
+
+// HPACK decoder for RESPONSE header blocks (RFC 7541): static + dynamic
+// table, multi-byte integers, Huffman-coded strings. One instance per
+// connection — the dynamic table persists across header blocks (grpc
+// servers add entries in the initial HEADERS that the trailers then
+// reference by index). Throws GrpcError on malformed input; unary_call
+// treats that as "headers unavailable" rather than failing the RPC.
+class HpackDecoder {
+public:
+  // `block` is one complete header block: the HEADERS payload (padding and
+  // priority stripped) plus any CONTINUATION payloads, concatenated.
+  std::vector<Header> decode_block(const std::string& block);
+
+private:
+  std::deque<Header> dynamic_;
+};
+
+// Huffman-decode one HPACK string literal (RFC 7541 §5.2 + Appendix B).
+// Public for tests; decode_block calls it for H-bit strings.
+std::string huffman_decode(const uint8_t* data, size_t len);
+
 // One unary call: sends `request_msg` (raw protobuf message bytes, framing
 // added here) to `method_path` (e.g.
 // "/opentelemetry.proto.collector.trace.v1.TraceService/Export") and returns
 // the concatenated response message bytes (gRPC frames unwrapped). Throws
-// GrpcError on connect/transport errors, stream reset, deadline, or a
-// request larger than kMaxRequestBytes.
+// GrpcError on connect/transport errors, stream reset, deadline, a request
+// larger than kMaxRequestBytes, or a non-OK gRPC status: response headers /
+// trailers are HPACK-decoded and a `grpc-status` other than 0 raises with
+// the status code and the (percent-decoded) `grpc-message` — without this a
+// collector rejecting an export (trailers-only response, no DATA) would be
+// indistinguishable from success, since ExportTraceServiceResponse is
+// legitimately empty.
 std::string unary_call(const Target& target, const std::string& method_path,
                        const std::string& request_msg, int timeout_ms);
 

@@ -166,6 +166,43 @@ PYBIND11_MODULE(_gpumon, m) {
         py::arg("request_msg"), py::arg("timeout_ms") = 5000,
         "Raw unary gRPC over h2c (native/common/grpc_client.cpp test surface)");
 
+  m.def("hpack_decode",
+        [](py::bytes block) {
+          std::string b = block;
+          grpcx::HpackDecoder d;
+          py::list out;
+          for (const auto& [k, v] : d.decode_block(b))
+            out.append(py::make_tuple(k, v));
+          return out;
+        },
+        py::arg("block"),
+        "Decode one HPACK header block (response-side decoder test surface)");
+
+  m.def("hpack_decode_blocks",
+        [](const std::vector<py::bytes>& blocks) {
+          grpcx::HpackDecoder d;  // dynamic table persists across blocks
+          py::list out;
+          for (const auto& blk : blocks) {
+            std::string b = blk;
+            py::list hs;
+            for (const auto& [k, v] : d.decode_block(b))
+              hs.append(py::make_tuple(k, v));
+            out.append(hs);
+          }
+          return out;
+        },
+        py::arg("blocks"),
+        "Decode successive header blocks sharing one dynamic table "
+        "(RFC 7541 Appendix C multi-request examples)");
+
+  m.def("huffman_decode",
+        [](py::bytes data) {
+          std::string d = data;
+          return py::bytes(grpcx::huffman_decode(
+              reinterpret_cast<const uint8_t*>(d.data()), d.size()));
+        },
+        py::arg("data"), "RFC 7541 Appendix B Huffman string decode");
+
   m.def("decode_list_response", [](py::bytes payload) {
     std::string data = payload;
     auto entries = decode_list_response(data);

@@ -87,3 +87,83 @@ def test_hpack_integer_coding_boundaries(gpumon, echo_server, n):
     assert len(method) == n
     assert gpumon.grpc_unary_call("127.0.0.1", echo_server["port"], method, b"k") == b"k"
     assert method in echo_server["received"]
+
+
+# ---------------------------------------------------------------------------
+# Response-header decoding (HPACK + grpc-status). Before this, the client
+# skipped response headers entirely, so a collector REJECTING an export
+# (trailers-only response: grpc-status != 0, no DATA) was indistinguishable
+# from success — ExportTraceServiceResponse is legitimately empty.
+# ---------------------------------------------------------------------------
+
+def test_hpack_decoder_rfc7541_appendix_c4_vectors(gpumon):
+    """RFC 7541 C.4.1-C.4.3: three consecutive Huffman-coded request header
+    blocks sharing one dynamic table — a cross-implementation check of the
+    Huffman table, the static table, and dynamic-table indexing (the hex is
+    the RFC's own, not produced by our encoder)."""
+    blocks = [
+        bytes.fromhex("828684418cf1e3c2e5f23a6ba0ab90f4ff"),
+        bytes.fromhex("828684be5886a8eb10649cbf"),
+        bytes.fromhex("828785bf408825a849e95ba97d7f8925a849e95bb8e8b4bf"),
+    ]
+    r = gpumon.hpack_decode_blocks(blocks)
+    assert r[0] == [(":method", "GET"), (":scheme", "http"), (":path", "/"),
+                    (":authority", "www.example.com")]
+    assert r[1] == r[0] + [("cache-control", "no-cache")]
+    assert r[2] == [(":method", "GET"), (":scheme", "https"),
+                    (":path", "/index.html"), (":authority", "www.example.com"),
+                    ("custom-key", "custom-value")]
+
+
+def test_huffman_decode_known_vector(gpumon):
+    assert gpumon.huffman_decode(
+        bytes.fromhex("f1e3c2e5f23a6ba0ab90f4ff")) == b"www.example.com"
+
+
+def test_huffman_decode_rejects_bad_padding(gpumon):
+    # 0x00 decodes '0' (code 00000) then leaves 3 zero bits of padding —
+    # RFC 7541 §5.2 requires padding be the all-ones EOS prefix
+    with pytest.raises(Exception):
+        gpumon.huffman_decode(bytes([0x00]))
+
+
+def test_hpack_decoder_rejects_out_of_range_index(gpumon):
+    with pytest.raises(Exception):
+        gpumon.hpack_decode(bytes([0x80 | 0x7F, 0x7F]))  # index far past tables
+
+
+def test_grpc_error_status_raises_with_decoded_message(gpumon):
+    """A real grpcio server aborting the RPC: trailers carry grpc-status 8
+    and a Huffman-coded, percent-encoded UTF-8 grpc-message. The client must
+    raise — not return an empty 'success' — and surface both."""
+    import grpc
+    from concurrent import futures
+
+    class Handler(grpc.GenericRpcHandler):
+        def service(self, hcd):
+            def unary_unary(request, context):
+                context.abort(grpc.StatusCode.RESOURCE_EXHAUSTED,
+                              "quota exceeded — try later")
+            return grpc.unary_unary_rpc_method_handler(
+                unary_unary, request_deserializer=None,
+                response_serializer=None)
+
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2),
+                         handlers=(Handler(),))
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    try:
+        with pytest.raises(gpumon.GrpcError) as ei:
+            gpumon.grpc_unary_call("127.0.0.1", port, "/t.S/M", b"x")
+        assert "grpc-status 8" in str(ei.value)
+        assert "quota exceeded — try later" in str(ei.value)
+    finally:
+        server.stop(grace=None)
+
+
+def test_grpc_ok_status_still_returns_payload(gpumon, echo_server):
+    """grpc-status 0 in trailers (now parsed) must not disturb the success
+    path."""
+    resp = gpumon.grpc_unary_call("127.0.0.1", echo_server["port"],
+                                  "/t.S/Ok", b"abc")
+    assert resp == b"cba"
