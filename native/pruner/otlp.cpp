@@ -316,7 +316,9 @@ void export_once() {
         try {
           grpc_export(s, "/opentelemetry.proto.collector.trace.v1.TraceService/Export",
                       std::move(msg));
-        } catch (const std::exception&) { /* collector away; drop batch */ }
+        } catch (const std::exception&) {
+          logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+        }
         start += take;
       }
       continue;
@@ -325,7 +327,9 @@ void export_once() {
       try {
         post_payload(s.endpoint + "/v1/traces", encode_spans_pb(s, batch),
                      "application/x-protobuf");
-      } catch (const std::exception&) { /* collector away; drop batch */ }
+      } catch (const std::exception&) {
+          logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+        }
       continue;
     }
     jsn::Value spans = jsn::Value::array();
@@ -351,7 +355,9 @@ void export_once() {
     payload["resourceSpans"] = jsn::Value(jsn::Array{rs});
     try {
       post_json(s.endpoint + "/v1/traces", payload);
-    } catch (const std::exception&) { /* collector away; drop batch */ }
+    } catch (const std::exception&) {
+          logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+        }
   }
 
   // ---- metrics: the counter registry (monotonic counters + gauges) ----
@@ -360,13 +366,17 @@ void export_once() {
     try {
       grpc_export(s, "/opentelemetry.proto.collector.metrics.v1.MetricsService/Export",
                   encode_metrics_pb(s, counters, now_unix_ns()));
-    } catch (const std::exception&) { /* collector away */ }
+    } catch (const std::exception&) {
+      logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+    }
   } else if (!counters.empty() && s.transport == Transport::HttpProtobuf) {
     try {
       post_payload(s.endpoint + "/v1/metrics",
                    encode_metrics_pb(s, counters, now_unix_ns()),
                    "application/x-protobuf");
-    } catch (const std::exception&) { /* collector away */ }
+    } catch (const std::exception&) {
+      logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+    }
   } else if (!counters.empty()) {
     uint64_t t = now_unix_ns();
     jsn::Value metrics = jsn::Value::array();
@@ -401,7 +411,9 @@ void export_once() {
     payload["resourceMetrics"] = jsn::Value(jsn::Array{rm});
     try {
       post_json(s.endpoint + "/v1/metrics", payload);
-    } catch (const std::exception&) { /* collector away */ }
+    } catch (const std::exception&) {
+      logx::counter_add("monotonic_counter.otlp_export_failures", 1);
+    }
   }
 }
 

@@ -10,6 +10,7 @@ span + counter surface, plus the three log formats.
 import json
 import os
 import subprocess
+import time
 
 import pytest
 
@@ -146,3 +147,63 @@ def test_log_format_pretty(pruner_bin, idle_cluster, fake_prom):
     # pretty format: level on its own styled segment, message indented
     assert any(l.startswith("    ") for l in r.stderr.splitlines())
     assert " INFO " in r.stderr
+
+
+def test_otlp_export_failure_counter(pruner_bin, idle_cluster, fake_prom):
+    """A collector that REJECTS every export (grpc-status RESOURCE_EXHAUSTED,
+    trailers-only response) must be counted as failures, not silently treated
+    as success — observable on the self-metrics endpoint since the OTLP
+    metrics path is the thing that's failing."""
+    grpc = pytest.importorskip("grpc")
+    import socket
+    import urllib.request
+    from concurrent import futures
+
+    class Rejecting(grpc.GenericRpcHandler):
+        def service(self, hcd):
+            def unary_unary(request, context):
+                context.abort(grpc.StatusCode.RESOURCE_EXHAUSTED, "no quota")
+            return grpc.unary_unary_rpc_method_handler(
+                unary_unary, request_deserializer=None,
+                response_serializer=None)
+
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2),
+                         handlers=(Rejecting(),))
+    gport = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    mport = s.getsockname()[1]
+    s.close()
+
+    env = dict(os.environ)
+    env["GPU_PRUNER_K8S_URL"] = idle_cluster.url
+    env["PROMETHEUS_TOKEN"] = "t"
+    env["OTEL_EXPORTER_OTLP_ENDPOINT"] = f"http://127.0.0.1:{gport}"
+    env["OTEL_EXPORTER_OTLP_PROTOCOL"] = "grpc"
+    env["OTEL_METRIC_EXPORT_INTERVAL"] = "200"
+    p = subprocess.Popen(
+        [pruner_bin, "--prometheus-url", fake_prom.url, "--daemon-mode",
+         "--check-interval", "1", "--run-mode", "dry-run",
+         "--metrics-port", str(mport)],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        failures = 0
+        for _ in range(50):
+            time.sleep(0.2)
+            try:
+                text = urllib.request.urlopen(
+                    f"http://127.0.0.1:{mport}/metrics", timeout=2).read().decode()
+            except OSError:
+                continue
+            for line in text.splitlines():
+                if line.startswith("gpu_pruner_otlp_export_failures_total"):
+                    failures = float(line.split()[-1])
+            if failures > 0:
+                break
+        assert failures > 0, "rejected exports were not counted as failures"
+    finally:
+        p.terminate()
+        p.wait(timeout=10)
+        server.stop(grace=None)
