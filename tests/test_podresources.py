@@ -220,3 +220,33 @@ def test_non_matching_ids_do_not_attribute(gpumon, tmp_path, monkeypatch, bogus_
         a = gpumon.Attributor()
         out = a.resolve_full(json.dumps([DEVICE0]))
     assert 0 not in out, (bogus_id, out)
+
+
+def test_kubelet_grpc_error_raises_loudly(gpumon, tmp_path):
+    """A kubelet returning a gRPC error (trailers-only: grpc-status != 0, no
+    DATA) must raise PodResourcesError carrying the status — before response
+    headers were decoded, it looked like an EMPTY pod list, silently
+    disabling GPU→pod attribution."""
+    import grpc
+    from concurrent import futures
+
+    class Failing(grpc.GenericRpcHandler):
+        def service(self, hcd):
+            def unary_unary(request, context):
+                context.abort(grpc.StatusCode.UNAVAILABLE, "kubelet shutting down")
+            return grpc.unary_unary_rpc_method_handler(
+                unary_unary, request_deserializer=None,
+                response_serializer=None)
+
+    sock = str(tmp_path / "kubelet.sock")
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2),
+                         handlers=(Failing(),))
+    server.add_insecure_port(f"unix:{sock}")
+    server.start()
+    try:
+        with pytest.raises(gpumon.PodResourcesError) as ei:
+            gpumon.list_pod_resources(sock)
+        assert "grpc-status 14" in str(ei.value)  # UNAVAILABLE
+        assert "kubelet shutting down" in str(ei.value)
+    finally:
+        server.stop(grace=None)
