@@ -95,7 +95,7 @@ def test_sigterm_graceful_shutdown(pruner_bin, fake_api, fake_prom):
 
 
 def test_self_metrics_endpoint(pruner_bin, fake_api, fake_prom):
-    """--metrics-port serves the six counters + /healthz (MI355X-native add)."""
+    """--metrics-port serves the counters + /healthz (MI355X-native add)."""
     import socket
     import urllib.request
 
