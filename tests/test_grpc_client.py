@@ -167,3 +167,17 @@ def test_grpc_ok_status_still_returns_payload(gpumon, echo_server):
     resp = gpumon.grpc_unary_call("127.0.0.1", echo_server["port"],
                                   "/t.S/Ok", b"abc")
     assert resp == b"cba"
+
+
+def test_hpack_decoder_remaining_opcodes(gpumon):
+    """Opcodes real servers emit that the C.4 vectors don't cover: dynamic
+    table size update (001xxxxx, must be consumed without emitting a
+    header), literal never-indexed (0001xxxx — how gRPC sends grpc-status),
+    and literal-without-indexing with a static name index."""
+    # size update then indexed static :method GET
+    assert gpumon.hpack_decode(bytes([0x20, 0x82])) == [(":method", "GET")]
+    # never-indexed literal name+value (the grpc-status trailer shape)
+    blk = bytes([0x10, 0x0B]) + b"grpc-status" + bytes([0x01]) + b"0"
+    assert gpumon.hpack_decode(blk) == [("grpc-status", "0")]
+    # literal w/o indexing, name from static table (8 = :status)
+    assert gpumon.hpack_decode(bytes([0x08, 0x03]) + b"200") == [(":status", "200")]
