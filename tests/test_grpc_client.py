@@ -181,3 +181,41 @@ def test_hpack_decoder_remaining_opcodes(gpumon):
     assert gpumon.hpack_decode(blk) == [("grpc-status", "0")]
     # literal w/o indexing, name from static table (8 = :status)
     assert gpumon.hpack_decode(bytes([0x08, 0x03]) + b"200") == [(":status", "200")]
+
+
+def test_hpack_literal_roundtrip_property(gpumon):
+    """Property check: an independent Python HPACK encoder (written to RFC
+    7541 §5.1/§6.2.3, in this test) round-trips through the C++ decoder for
+    arbitrary ASCII headers — exercises string-length integer boundaries
+    (127/128, multi-byte continuations) the fixed vectors can't sweep."""
+    hypothesis = pytest.importorskip("hypothesis")
+    from hypothesis import given, settings, strategies as st
+
+    def enc_int(prefix_bits, first_bits, v):
+        max_p = (1 << prefix_bits) - 1
+        if v < max_p:
+            return bytes([first_bits | v])
+        out = bytearray([first_bits | max_p])
+        v -= max_p
+        while v >= 128:
+            out.append((v & 0x7F) | 0x80)
+            v >>= 7
+        out.append(v)
+        return bytes(out)
+
+    def enc_literal(name, value):  # never-indexed, literal name, raw strings
+        return (enc_int(4, 0x10, 0)
+                + enc_int(7, 0x00, len(name)) + name
+                + enc_int(7, 0x00, len(value)) + value)
+
+    ascii_txt = st.text(
+        alphabet=st.characters(min_codepoint=0x20, max_codepoint=0x7E),
+        max_size=300)
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.lists(st.tuples(ascii_txt, ascii_txt), max_size=8))
+    def roundtrip(headers):
+        block = b"".join(enc_literal(n.encode(), v.encode()) for n, v in headers)
+        assert gpumon.hpack_decode(block) == headers
+
+    roundtrip()
