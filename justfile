@@ -21,6 +21,8 @@ test: build
 test-all: test
     make -C native asan
     make -C native tsan
+    make -C native tsan-bin
+    python3 -m pytest tests/test_tsan_binary.py -q
 
 # GPU tier (requires an MI355X; the driver runs this via gpurun)
 test-gpu:
